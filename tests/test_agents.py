@@ -1,0 +1,101 @@
+"""Agent-level tests: SAC/TD3/DDPG learn steps, checkpoints, polyak."""
+
+import numpy as np
+import pytest
+import torch
+
+from smartcal_amd.rl import ddpg, sac, td3
+
+CPU = torch.device("cpu")
+
+
+def _fake_obs(n_state, n=8):
+    # state layout: eig (n) + A (n_state - n)
+    return {"eig": torch.randn(n), "A": torch.randn(n_state - n)}
+
+
+def _fill_and_learn(agent, n_state, steps=3):
+    obs = _fake_obs(n_state)
+    for _ in range(agent.batch_size + 2):
+        a = agent.choose_action(obs)
+        obs2 = _fake_obs(n_state)
+        agent.store_transition(obs, a, float(np.random.randn()), obs2,
+                               False, np.zeros_like(a))
+        obs = obs2
+    before = agent.actor_fp.flat.clone()
+    for _ in range(steps):
+        agent.learn()
+    return before
+
+
+def test_sac_learn_updates_params(tmp_path):
+    agent = sac.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=[24],
+                      batch_size=8, n_actions=2, max_mem_size=64,
+                      tau=0.005, reward_scale=2, alpha=0.03, device=CPU,
+                      checkpoint_dir=str(tmp_path))
+    before = _fill_and_learn(agent, 24)
+    assert not torch.allclose(before, agent.actor_fp.flat)
+    # target critics track online critics
+    assert not torch.allclose(agent.target_critic_1_fp.flat,
+                              agent.critic_1_fp.flat)
+    agent.save_models()
+    agent2 = sac.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=[24],
+                       batch_size=8, n_actions=2, max_mem_size=64,
+                       tau=0.005, reward_scale=2, alpha=0.03, device=CPU,
+                       checkpoint_dir=str(tmp_path))
+    agent2.load_models()
+    assert torch.allclose(agent2.actor_fp.flat, agent.actor_fp.flat)
+    # hard target sync after load (tau=1)
+    assert torch.allclose(agent2.target_critic_1_fp.flat,
+                          agent2.critic_1_fp.flat)
+
+
+def test_sac_prioritized_and_hint(tmp_path):
+    agent = sac.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=[24],
+                      batch_size=8, n_actions=2, max_mem_size=64,
+                      tau=0.005, reward_scale=2, alpha=0.03,
+                      prioritized=True, use_hint=True, device=CPU,
+                      checkpoint_dir=str(tmp_path))
+    _fill_and_learn(agent, 24, steps=11)
+    assert agent.learn_counter == 11
+    assert float(agent.rho) >= 0.0
+
+
+def test_td3_learn_and_warmup():
+    agent = td3.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=[24],
+                      batch_size=8, n_actions=2, max_mem_size=64,
+                      tau=0.005, warmup=5, noise=0.1, prioritized=True,
+                      device=CPU)
+    obs = _fake_obs(24)
+    acts = [agent.choose_action(obs) for _ in range(6)]
+    assert all(a.shape == (2,) for a in acts)
+    before = _fill_and_learn(agent, 24, steps=4)
+    assert not torch.allclose(before, agent.actor_fp.flat)
+
+
+def test_td3_hint_admm():
+    agent = td3.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=[24],
+                      batch_size=8, n_actions=2, max_mem_size=64,
+                      tau=0.005, warmup=0, noise=0.1, use_hint=True,
+                      device=CPU)
+    _fill_and_learn(agent, 24, steps=4)
+
+
+def test_ddpg_learn():
+    agent = ddpg.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=[24],
+                       batch_size=8, n_actions=2, max_mem_size=64,
+                       tau=0.005, device=CPU)
+    before = _fill_and_learn(agent, 24, steps=3)
+    assert not torch.allclose(before, agent.actor_fp.flat)
+
+
+def test_polyak_flat_matches_reference_rule():
+    agent = ddpg.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=[24],
+                       batch_size=8, n_actions=2, max_mem_size=64,
+                       tau=0.3, device=CPU)
+    online = agent.critic_fp.flat.clone()
+    target0 = torch.randn_like(online)
+    agent.target_critic_fp.flat.copy_(target0)
+    agent.update_network_parameters()  # tau = 0.3
+    expected = 0.3 * online + 0.7 * target0
+    assert torch.allclose(agent.target_critic_fp.flat, expected, atol=1e-6)

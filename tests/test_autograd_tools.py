@@ -1,0 +1,77 @@
+"""autograd_tools: Jacobian / HVP / iHVP / influence_matrix."""
+
+import torch
+
+from smartcal_amd import autograd_tools as at
+
+
+def test_jacobian_matches_torch():
+    A = torch.randn(6, 4)
+    x = torch.randn(4, requires_grad=True)
+    y = A @ x
+    jac = at.jacobian(y, x)
+    assert torch.allclose(jac, A, atol=1e-6)
+
+
+def test_gradient_vjp():
+    x = torch.randn(5, requires_grad=True)
+    y = (x ** 2).sum()
+    g = at.gradient(y, x)
+    assert torch.allclose(g, 2 * x, atol=1e-6)
+
+
+class TinyNet(torch.nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.l1 = torch.nn.Linear(3, 4)
+        self.l2 = torch.nn.Linear(4, 2)
+
+    def forward(self, x):
+        return self.l2(torch.tanh(self.l1(x)))
+
+
+def test_hessian_vec_prod_quadratic():
+    # for loss = ||W x - y||^2 / n wrt W (linear model), Hessian is constant;
+    # check HVP against finite differences of the gradient.
+    torch.manual_seed(0)
+    net = torch.nn.Linear(3, 2, bias=False)
+    x = torch.randn(7, 3)
+    y = torch.randn(7, 2)
+    crit = torch.nn.MSELoss()
+    v = torch.randn(sum(p.numel() for p in net.parameters()))
+    hv = at.hessian_vec_prod(net, crit, x, y, v)
+
+    eps = 1e-3
+
+    def flat_grad_at(shift):
+        with torch.no_grad():
+            offset = 0
+            for p in net.parameters():
+                n = p.numel()
+                p.add_(shift[offset:offset + n].view_as(p))
+                offset += n
+        net.zero_grad()
+        crit(net(x), y).backward()
+        g = torch.cat([p.grad.reshape(-1).clone() for p in net.parameters()])
+        with torch.no_grad():
+            offset = 0
+            for p in net.parameters():
+                n = p.numel()
+                p.sub_(shift[offset:offset + n].view_as(p))
+                offset += n
+        return g
+
+    g_plus = flat_grad_at(eps * v)
+    g_minus = flat_grad_at(-eps * v)
+    hv_fd = (g_plus - g_minus) / (2 * eps)
+    assert torch.allclose(hv, hv_fd, rtol=1e-2, atol=1e-3)
+
+
+def test_influence_matrix_runs():
+    torch.manual_seed(0)
+    net = TinyNet()
+    x = torch.randn(3)
+    y = torch.randn(2)
+    If = at.influence_matrix(net, x, y)
+    assert If.shape == (2, 3)
+    assert torch.isfinite(If).all()
