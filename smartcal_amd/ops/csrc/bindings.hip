@@ -1,0 +1,241 @@
+// Python bindings for the smartcal_amd HIP/CDNA4 kernels (gfx950 only).
+// Explicit HIP APIs throughout — no CUDA-compat shims.
+
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+#include <torch/extension.h>
+
+#include <tuple>
+
+#include "common.h"
+
+#define HMAX 7
+
+// ---- kernel declarations (definitions in the sibling .hip files) ----
+extern "C" __global__ void fused_linear_fwd_kernel(
+    const float*, const float*, const float*, const float*, const float*,
+    float*, float*, float*, int, int, int, int, int);
+extern "C" __global__ void ln_act_bwd_kernel(const float*, const float*,
+                                             const float*, const float*,
+                                             const float*, float*, float*,
+                                             float*, int, int, int, int);
+extern "C" __global__ void gemm_f32_nn_kernel(const float*, const float*,
+                                              float*, int, int, int);
+extern "C" __global__ void gemm_f32_tn_kernel(const float*, const float*,
+                                              float*, int, int, int);
+extern "C" __global__ void colsum_kernel(const float*, float*, int, int);
+extern "C" __global__ void tanh_gauss_fwd_kernel(const float*, const float*,
+                                                 const float*, float*, float*,
+                                                 float*, float, int, int);
+extern "C" __global__ void tanh_gauss_bwd_kernel(const float*, const float*,
+                                                 const float*, const float*,
+                                                 const float*, float*, float*,
+                                                 float, int, int);
+extern "C" __global__ void fused_adam_kernel(float*, const float*, float*,
+                                             float*, float, float, float,
+                                             float, float, float, long);
+extern "C" __global__ void enet_lbfgs_solve_kernel(
+    const float*, const float*, const float*, float*, float*, float*, int*,
+    int, int, int, int, int, int);
+extern "C" __global__ void enet_influence_kernel(
+    const float*, const float*, const float*, const float*, const float*,
+    const int*, const float*, float*, float*, int, int, int);
+
+namespace {
+
+inline void check_f32(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on the GPU");
+  TORCH_CHECK(t.scalar_type() == at::kFloat, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+inline hipStream_t stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_fwd(
+    const at::Tensor& x, const at::Tensor& W,
+    const c10::optional<at::Tensor>& bias,
+    const c10::optional<at::Tensor>& gamma,
+    const c10::optional<at::Tensor>& beta, int64_t act, bool with_ln) {
+  check_f32(x, "x");
+  check_f32(W, "W");
+  const int B = x.size(0), K = x.size(1), N = W.size(0);
+  TORCH_CHECK(W.size(1) == K, "W/K mismatch");
+  TORCH_CHECK(N <= 576, "fused_linear: N>576 unsupported (extend NT_MAX)");
+  TORCH_CHECK(!with_ln || gamma.has_value(), "LN requires gamma");
+  auto y = at::empty({B, N}, x.options());
+  auto zhat = with_ln ? at::empty({B, N}, x.options())
+                      : at::empty({0}, x.options());
+  auto rstd = with_ln ? at::empty({B}, x.options())
+                      : at::empty({0}, x.options());
+  dim3 grid((B + 15) / 16);
+  hipLaunchKernelGGL(fused_linear_fwd_kernel, grid, dim3(256), 0, stream(),
+                     x.data_ptr<float>(), W.data_ptr<float>(),
+                     bias ? bias->data_ptr<float>() : nullptr,
+                     gamma ? gamma->data_ptr<float>() : nullptr,
+                     beta ? beta->data_ptr<float>() : nullptr,
+                     y.data_ptr<float>(),
+                     with_ln ? zhat.data_ptr<float>() : nullptr,
+                     with_ln ? rstd.data_ptr<float>() : nullptr, B, K, N,
+                     (int)act, with_ln ? 1 : 0);
+  return {y, zhat, rstd};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_bwd_dz(
+    const at::Tensor& dy, const at::Tensor& y, const at::Tensor& zhat,
+    const at::Tensor& rstd, const c10::optional<at::Tensor>& gamma,
+    int64_t act, bool with_ln) {
+  check_f32(dy, "dy");
+  const int B = dy.size(0), N = dy.size(1);
+  auto dz = at::empty({B, N}, dy.options());
+  auto dgamma = at::zeros({N}, dy.options());
+  auto dbeta = at::zeros({N}, dy.options());
+  hipLaunchKernelGGL(ln_act_bwd_kernel, dim3(B), dim3(256), 0, stream(),
+                     dy.data_ptr<float>(), y.data_ptr<float>(),
+                     with_ln ? zhat.data_ptr<float>() : nullptr,
+                     with_ln ? rstd.data_ptr<float>() : nullptr,
+                     gamma ? gamma->data_ptr<float>() : nullptr,
+                     dz.data_ptr<float>(), dgamma.data_ptr<float>(),
+                     dbeta.data_ptr<float>(), B, N, (int)act,
+                     with_ln ? 1 : 0);
+  return {dz, dgamma, dbeta};
+}
+
+at::Tensor mfma_gemm_nn(const at::Tensor& A, const at::Tensor& B) {
+  check_f32(A, "A");
+  check_f32(B, "B");
+  const int M = A.size(0), K = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == K, "gemm_nn shape mismatch");
+  auto C = at::empty({M, N}, A.options());
+  dim3 grid((M + 15) / 16, (N + 63) / 64);
+  hipLaunchKernelGGL(gemm_f32_nn_kernel, grid, dim3(256), 0, stream(),
+                     A.data_ptr<float>(), B.data_ptr<float>(),
+                     C.data_ptr<float>(), M, K, N);
+  return C;
+}
+
+std::tuple<at::Tensor, at::Tensor> mfma_gemm_tn_bias(const at::Tensor& dz,
+                                                     const at::Tensor& x) {
+  check_f32(dz, "dz");
+  check_f32(x, "x");
+  const int Bb = dz.size(0), N = dz.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == Bb, "gemm_tn batch mismatch");
+  auto dW = at::empty({N, K}, dz.options());
+  auto db = at::empty({N}, dz.options());
+  dim3 grid((N + 15) / 16, (K + 63) / 64);
+  hipLaunchKernelGGL(gemm_f32_tn_kernel, grid, dim3(256), 0, stream(),
+                     dz.data_ptr<float>(), x.data_ptr<float>(),
+                     dW.data_ptr<float>(), N, Bb, K);
+  hipLaunchKernelGGL(colsum_kernel, dim3((N + 255) / 256), dim3(256), 0,
+                     stream(), dz.data_ptr<float>(), db.data_ptr<float>(),
+                     Bb, N);
+  return {dW, db};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> tanh_gauss_fwd(
+    const at::Tensor& mu, const at::Tensor& logsigma, const at::Tensor& eps,
+    double max_action) {
+  check_f32(mu, "mu");
+  const int B = mu.size(0), A = mu.size(1);
+  auto action = at::empty_like(mu);
+  auto at_ = at::empty_like(mu);
+  auto logp = at::empty({B, 1}, mu.options());
+  hipLaunchKernelGGL(tanh_gauss_fwd_kernel, dim3(B), dim3(64), 0, stream(),
+                     mu.data_ptr<float>(), logsigma.data_ptr<float>(),
+                     eps.data_ptr<float>(), action.data_ptr<float>(),
+                     logp.data_ptr<float>(), at_.data_ptr<float>(),
+                     (float)max_action, B, A);
+  return {action, logp, at_};
+}
+
+std::tuple<at::Tensor, at::Tensor> tanh_gauss_bwd(
+    const at::Tensor& dact, const at::Tensor& dlogp,
+    const at::Tensor& logsigma, const at::Tensor& eps, const at::Tensor& at_,
+    double max_action) {
+  check_f32(dact, "dact");
+  const int B = dact.size(0), A = dact.size(1);
+  auto dmu = at::empty_like(dact);
+  auto dls = at::empty_like(dact);
+  const long n = (long)B * A;
+  hipLaunchKernelGGL(tanh_gauss_bwd_kernel, dim3((n + 255) / 256), dim3(256),
+                     0, stream(), dact.data_ptr<float>(),
+                     dlogp.data_ptr<float>(), logsigma.data_ptr<float>(),
+                     eps.data_ptr<float>(), at_.data_ptr<float>(),
+                     dmu.data_ptr<float>(), dls.data_ptr<float>(),
+                     (float)max_action, B, A);
+  return {dmu, dls};
+}
+
+void fused_adam(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
+                at::Tensor& v, double lr, double b1, double b2, double eps,
+                int64_t t) {
+  check_f32(p, "p");
+  const long n = p.numel();
+  const float bc1 = 1.f - powf((float)b1, (float)t);
+  const float bc2 = 1.f - powf((float)b2, (float)t);
+  hipLaunchKernelGGL(fused_adam_kernel, dim3((n + 255) / 256), dim3(256), 0,
+                     stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
+                     (float)b1, (float)b2, (float)eps, bc1, bc2, n);
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> enet_lbfgs_solve(
+    const at::Tensor& A, const at::Tensor& y, const at::Tensor& rho,
+    int64_t epochs, int64_t max_iter, int64_t history) {
+  check_f32(A, "A");
+  check_f32(y, "y");
+  check_f32(rho, "rho");
+  const int E = A.size(0), N = A.size(1), M = A.size(2);
+  TORCH_CHECK(N <= 32 && M <= 32, "enet solver supports N,M <= 32");
+  TORCH_CHECK(history <= HMAX, "history <= 7");
+  auto x = at::empty({E, M}, A.options());
+  auto Y = at::empty({E, (long)HMAX, M}, A.options());
+  auto S = at::empty({E, (long)HMAX, M}, A.options());
+  auto nh = at::empty({E}, A.options().dtype(at::kInt));
+  const int lds_floats = N * M + N + 5 * M + N + 2 * HMAX * M + 2 * M
+                         + 2 * HMAX;
+  hipLaunchKernelGGL(enet_lbfgs_solve_kernel, dim3(E), dim3(64),
+                     lds_floats * sizeof(float), stream(),
+                     A.data_ptr<float>(), y.data_ptr<float>(),
+                     rho.data_ptr<float>(), x.data_ptr<float>(),
+                     Y.data_ptr<float>(), S.data_ptr<float>(),
+                     nh.data_ptr<int>(), E, N, M, (int)epochs, (int)max_iter,
+                     (int)history);
+  return {x, Y, S, nh};
+}
+
+std::tuple<at::Tensor, at::Tensor> enet_influence(
+    const at::Tensor& A, const at::Tensor& y, const at::Tensor& x,
+    const at::Tensor& Y, const at::Tensor& S, const at::Tensor& nh,
+    const at::Tensor& penalty) {
+  check_f32(A, "A");
+  const int E = A.size(0), N = A.size(1), M = A.size(2);
+  TORCH_CHECK(N <= 32 && M <= 32, "enet influence supports N,M <= 32");
+  auto EE = at::empty({E, N}, A.options());
+  auto reward = at::empty({E}, A.options());
+  const int lds_floats = N * M + M * N + N * N + 2 * HMAX * M + HMAX * N + N
+                         + M + N + HMAX;
+  hipLaunchKernelGGL(enet_influence_kernel, dim3(E), dim3(64),
+                     lds_floats * sizeof(float), stream(),
+                     A.data_ptr<float>(), y.data_ptr<float>(),
+                     x.data_ptr<float>(), Y.data_ptr<float>(),
+                     S.data_ptr<float>(), nh.data_ptr<int>(),
+                     penalty.data_ptr<float>(), EE.data_ptr<float>(),
+                     reward.data_ptr<float>(), E, N, M);
+  return {EE, reward};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_linear_fwd", &fused_linear_fwd);
+  m.def("fused_linear_bwd_dz", &fused_linear_bwd_dz);
+  m.def("mfma_gemm_nn", &mfma_gemm_nn);
+  m.def("mfma_gemm_tn_bias", &mfma_gemm_tn_bias);
+  m.def("tanh_gauss_fwd", &tanh_gauss_fwd);
+  m.def("tanh_gauss_bwd", &tanh_gauss_bwd);
+  m.def("fused_adam", &fused_adam);
+  m.def("enet_lbfgs_solve", &enet_lbfgs_solve);
+  m.def("enet_influence", &enet_influence);
+}

@@ -1,0 +1,223 @@
+// Fused Linear(+LayerNorm)(+activation) forward and backward — the RL
+// actor/critic MLP hot op (reference composes
+// Linear -> LayerNorm -> ELU as separate torch ops, enet_sac.py:436-444).
+//
+// Forward kernel: one workgroup (4 waves, 256 threads) owns a 16-row slab of
+// the batch and ALL output columns (so the LayerNorm row statistics never
+// leave the workgroup). The GEMM runs on v_mfma_f32_16x16x4_f32 (exact f32,
+// §3 of the CDNA4 guide) with the X tile staged in LDS (+1 pad, conflict
+// free) and W fragments streamed from L2 (W is re-read only gridDim.x =
+// ceil(B/16) times; RL batches are 1..256 so W stays L2-resident). The
+// epilogue fuses bias + LayerNorm (shfl_xor 16-lane row reductions + one
+// LDS cross-wave combine) + activation, and stores y, zhat, rstd for the
+// backward pass. No separate normalization kernel, no extra HBM round trip.
+//
+// Backward: (1) ln_act_bwd_kernel — act'(y) + LayerNorm backward (row
+// reductions in-workgroup) + dgamma/dbeta column partials; (2)/(3) are MFMA
+// GEMMs in gemm_f32.hip.
+
+#include "common.h"
+
+#define NT_MAX 9           // max 16-col tiles per wave => N <= 4*NT_MAX*16
+#define BK 64              // K-tile staged in LDS
+
+extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
+    const float* __restrict__ X,    // (B, K)
+    const float* __restrict__ W,    // (N, K) row-major
+    const float* __restrict__ bias, // (N) or null
+    const float* __restrict__ gamma,// (N) or null (=> no LN)
+    const float* __restrict__ beta, // (N)
+    float* __restrict__ Y,          // (B, N)
+    float* __restrict__ ZHAT,       // (B, N) (LN only)
+    float* __restrict__ RSTD,       // (B,)   (LN only)
+    int B, int K, int N, int act, int with_ln) {
+  __shared__ float smem[16 * (BK + 1) + 4 * 16 * 2 + 16 * 2];
+  float* xs = smem;                      // [16][BK+1]
+  float* rowstat = smem + 16 * (BK + 1); // [4 waves][16 rows][2]
+  float* rowmv = rowstat + 4 * 16 * 2;   // [16][2] mean, rstd
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;          // 0..3
+  const int row0 = blockIdx.x * 16;  // first batch row of this block
+
+  const int ntiles = (N + 15) >> 4;  // 16-col tiles over N
+
+  f32x4 acc[NT_MAX];
+#pragma unroll
+  for (int t = 0; t < NT_MAX; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  for (int kk = 0; kk < K; kk += BK) {
+    // stage X[row0..row0+15][kk..kk+BK) into LDS (zero-padded)
+    for (int idx = tid; idx < 16 * BK; idx += 256) {
+      int r = idx / BK, c = idx % BK;
+      int gr = row0 + r, gc = kk + c;
+      xs[r * (BK + 1) + c] =
+          (gr < B && gc < K) ? X[(long)gr * K + gc] : 0.f;
+    }
+    __syncthreads();
+    const int kmax = min(BK, K - kk);
+#pragma unroll
+    for (int t = 0; t < NT_MAX; ++t) {
+      const int ct = wave + 4 * t;
+      if (ct < ntiles) {
+        const int col = ct * 16 + l15;
+        const bool colv = col < N;
+        const float* wrow = W + (long)col * K + kk;
+        for (int k4 = 0; k4 < BK / 4; ++k4) {
+          // MFMA is a whole-wave op: never predicate it per-lane. Edge
+          // tiles contribute zeros through the operands instead (X tile is
+          // zero-padded in LDS; W reads are masked here).
+          const int k = k4 * 4 + l4;
+          float a = xs[l15 * (BK + 1) + k];
+          float b = (colv && k < kmax) ? wrow[k] : 0.f;
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[t], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + LN + act ----
+  // acc[t][r] holds z[row = row0 + l4*4 + r][col = (wave+4t)*16 + l15]
+  float zrow[NT_MAX][4];
+  float psum[4] = {0.f, 0.f, 0.f, 0.f}, psq[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int t = 0; t < NT_MAX; ++t) {
+    const int ct = wave + 4 * t;
+    const int col = ct * 16 + l15;
+    const bool colv = (ct < ntiles) && (col < N);
+    const float bv = (colv && bias) ? bias[col] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float z = colv ? acc[t][r] + bv : 0.f;
+      zrow[t][r] = z;
+      psum[r] += z;
+      psq[r] += z * z;
+    }
+  }
+
+  if (with_ln) {
+    // 16-lane reduction: lanes sharing l4 cover 16 distinct cols
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      psum[r] = group16_sum(psum[r]);
+      psq[r] = group16_sum(psq[r]);
+    }
+    if (l15 == 0) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        rowstat[(wave * 16 + l4 * 4 + r) * 2 + 0] = psum[r];
+        rowstat[(wave * 16 + l4 * 4 + r) * 2 + 1] = psq[r];
+      }
+    }
+    __syncthreads();
+    if (tid < 16) {
+      float s = 0.f, q = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        s += rowstat[(w * 16 + tid) * 2 + 0];
+        q += rowstat[(w * 16 + tid) * 2 + 1];
+      }
+      const float mean = s / N;
+      float var = q / N - mean * mean;
+      const float rstd = rsqrtf(fmaxf(var, 0.f) + 1e-5f);
+      rowmv[tid * 2 + 0] = mean;
+      rowmv[tid * 2 + 1] = rstd;
+      const int grow = row0 + tid;
+      if (grow < B && RSTD) RSTD[grow] = rstd;
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int lrow = l4 * 4 + r;
+    const int grow = row0 + lrow;
+    if (grow >= B) continue;
+    float mean = 0.f, rstd = 1.f;
+    if (with_ln) {
+      mean = rowmv[lrow * 2 + 0];
+      rstd = rowmv[lrow * 2 + 1];
+    }
+#pragma unroll
+    for (int t = 0; t < NT_MAX; ++t) {
+      const int ct = wave + 4 * t;
+      const int col = ct * 16 + l15;
+      if (ct < ntiles && col < N) {
+        float z = zrow[t][r];
+        float out;
+        if (with_ln) {
+          const float zh = (z - mean) * rstd;
+          ZHAT[(long)grow * N + col] = zh;
+          out = apply_act(gamma[col] * zh + beta[col], act);
+        } else {
+          out = apply_act(z, act);
+        }
+        Y[(long)grow * N + col] = out;
+      }
+    }
+  }
+}
+
+// LayerNorm+activation backward: dz from dy, plus dgamma/dbeta column sums.
+// One workgroup per batch row (block reductions over the N columns).
+extern "C" __global__ __launch_bounds__(256) void ln_act_bwd_kernel(
+    const float* __restrict__ DY,   // (B, N)
+    const float* __restrict__ Yv,   // (B, N) activation output
+    const float* __restrict__ ZHAT, // (B, N)
+    const float* __restrict__ RSTD, // (B,)
+    const float* __restrict__ gamma,// (N)
+    float* __restrict__ DZ,         // (B, N)
+    float* __restrict__ DGAMMA,     // (N) pre-zeroed, atomically accumulated
+    float* __restrict__ DBETA,      // (N)
+    int B, int N, int act, int with_ln) {
+  __shared__ float red[2 * 4];  // per-wave partial sums (m1, m2)
+  const int row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+
+  if (!with_ln) {
+    for (int c = tid; c < N; c += blockDim.x) {
+      const long i = (long)row * N + c;
+      DZ[i] = DY[i] * act_grad_from_y(Yv[i], act);
+    }
+    return;
+  }
+
+  float m1 = 0.f, m2 = 0.f;
+  for (int c = tid; c < N; c += blockDim.x) {
+    const long i = (long)row * N + c;
+    const float dh = DY[i] * act_grad_from_y(Yv[i], act);
+    const float zh = ZHAT[i];
+    const float dzh = dh * gamma[c];
+    m1 += dzh;
+    m2 += dzh * zh;
+    atomicAdd(&DGAMMA[c], dh * zh);
+    atomicAdd(&DBETA[c], dh);
+  }
+  m1 = wave_sum(m1);
+  m2 = wave_sum(m2);
+  if ((tid & 63) == 0) {
+    red[wave * 2 + 0] = m1;
+    red[wave * 2 + 1] = m2;
+  }
+  __syncthreads();
+  float M1 = 0.f, M2 = 0.f;
+  const int nw = (blockDim.x + 63) / 64;
+  for (int w = 0; w < nw; ++w) {
+    M1 += red[w * 2 + 0];
+    M2 += red[w * 2 + 1];
+  }
+  M1 /= N;
+  M2 /= N;
+  const float rstd = RSTD[row];
+  for (int c = tid; c < N; c += blockDim.x) {
+    const long i = (long)row * N + c;
+    const float dh = DY[i] * act_grad_from_y(Yv[i], act);
+    const float dzh = dh * gamma[c];
+    DZ[i] = rstd * (dzh - M1 - ZHAT[i] * M2);
+  }
+}

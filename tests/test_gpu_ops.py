@@ -1,0 +1,234 @@
+"""GPU numerics tests: every HIP kernel against its plain-PyTorch fp32
+oracle (the CPU implementations in smartcal_amd.ops)."""
+
+import numpy as np
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    import smartcal_amd.ops as ops
+    DEV = torch.device("cuda:0")
+else:
+    DEV = None
+
+
+@pytest.fixture(autouse=True)
+def _need_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    assert ops.have_hip(), "HIP extension must be loaded on a GPU box"
+
+
+def test_fused_linear_fwd_matches_cpu():
+    for (B, K, N) in [(64, 420, 512), (1, 420, 512), (64, 512, 256),
+                      (64, 128, 4), (3, 33, 17), (64, 320, 1)]:
+        x = torch.randn(B, K)
+        W = torch.randn(N, K) * 0.05
+        b = torch.randn(N) * 0.1
+        g = torch.rand(N) + 0.5
+        be = torch.randn(N) * 0.1
+        ref = F.elu(F.layer_norm(F.linear(x, W, b), (N,), g, be))
+        y, zhat, rstd = ops.ext().fused_linear_fwd(
+            x.to(DEV), W.to(DEV), b.to(DEV), g.to(DEV), be.to(DEV), 1, True)
+        assert torch.allclose(y.cpu(), ref, atol=2e-4), \
+            f"shape {(B, K, N)} max err {(y.cpu()-ref).abs().max()}"
+        # no-LN + no-act path
+        ref2 = F.linear(x, W, b)
+        y2, _, _ = ops.ext().fused_linear_fwd(
+            x.to(DEV), W.to(DEV), b.to(DEV), None, None, 0, False)
+        assert torch.allclose(y2.cpu(), ref2, atol=2e-4)
+
+
+def test_fused_linear_backward_matches_cpu():
+    torch.manual_seed(0)
+    B, K, N = 64, 420, 512
+    x = torch.randn(B, K)
+    W = torch.randn(N, K) * 0.05
+    b = torch.randn(N) * 0.1
+    g = torch.rand(N) + 0.5
+    be = torch.randn(N) * 0.1
+
+    # CPU oracle grads
+    xc = x.clone().requires_grad_(True)
+    Wc = W.clone().requires_grad_(True)
+    bc = b.clone().requires_grad_(True)
+    gc = g.clone().requires_grad_(True)
+    bec = be.clone().requires_grad_(True)
+    out = F.elu(F.layer_norm(F.linear(xc, Wc, bc), (N,), gc, bec))
+    loss = (out * torch.arange(N).float() / N).sum()
+    loss.backward()
+
+    from smartcal_amd.ops.linear import fused_linear
+    xg = x.clone().to(DEV).requires_grad_(True)
+    Wg = W.clone().to(DEV).requires_grad_(True)
+    bg = b.clone().to(DEV).requires_grad_(True)
+    gg = g.clone().to(DEV).requires_grad_(True)
+    beg = be.clone().to(DEV).requires_grad_(True)
+    outg = fused_linear(xg, Wg, bg, gg, beg, act="elu")
+    assert torch.allclose(outg.detach().cpu(), out.detach(), atol=2e-4)
+    lossg = (outg * (torch.arange(N, device=DEV).float() / N)).sum()
+    lossg.backward()
+
+    for cpu_t, gpu_t, name, tol in [
+            (xc.grad, xg.grad, "dx", 5e-4),
+            (Wc.grad, Wg.grad, "dW", 5e-4),
+            (bc.grad, bg.grad, "db", 5e-4),
+            (gc.grad, gg.grad, "dgamma", 5e-4),
+            (bec.grad, beg.grad, "dbeta", 5e-4)]:
+        err = (gpu_t.cpu() - cpu_t).abs().max()
+        denom = cpu_t.abs().max().clamp_min(1.0)
+        assert err / denom < tol, f"{name}: rel err {err/denom}"
+
+
+def test_mfma_gemms_match_torch():
+    A = torch.randn(64, 512)
+    B = torch.randn(512, 420)
+    C = ops.ext().mfma_gemm_nn(A.to(DEV), B.to(DEV))
+    ref = A @ B
+    assert torch.allclose(C.cpu(), ref, atol=ref.abs().max() * 1e-5 + 1e-4)
+
+    dz = torch.randn(64, 512)
+    x = torch.randn(64, 420)
+    dW, db = ops.ext().mfma_gemm_tn_bias(dz.to(DEV), x.to(DEV))
+    refW = dz.t() @ x
+    assert torch.allclose(dW.cpu(), refW,
+                          atol=refW.abs().max() * 1e-5 + 1e-4)
+    assert torch.allclose(db.cpu(), dz.sum(0), atol=1e-3)
+
+
+def test_tanh_gauss_matches_cpu():
+    from smartcal_amd.ops.sampling import tanh_gauss_sample
+    B, A = 64, 2
+    mu = torch.randn(B, A)
+    ls = torch.randn(B, A) * 0.3
+    eps = torch.randn(B, A)
+
+    a_cpu, lp_cpu = tanh_gauss_sample(
+        mu.clone().requires_grad_(True), ls.clone().requires_grad_(True),
+        1.0, True, eps)
+    mug = mu.clone().to(DEV).requires_grad_(True)
+    lsg = ls.clone().to(DEV).requires_grad_(True)
+    a_gpu, lp_gpu = tanh_gauss_sample(mug, lsg, 1.0, True, eps.to(DEV))
+    assert torch.allclose(a_gpu.detach().cpu(), a_cpu.detach(), atol=1e-5)
+    assert torch.allclose(lp_gpu.detach().cpu(), lp_cpu.detach(), atol=1e-4)
+
+    (a_gpu.sum() + lp_gpu.sum()).backward()
+    muc = mu.clone().requires_grad_(True)
+    lsc = ls.clone().requires_grad_(True)
+    a2, lp2 = tanh_gauss_sample(muc, lsc, 1.0, True, eps)
+    (a2.sum() + lp2.sum()).backward()
+    assert torch.allclose(mug.grad.cpu(), muc.grad, atol=1e-4)
+    assert torch.allclose(lsg.grad.cpu(), lsc.grad, atol=1e-4)
+
+
+def test_fused_adam_matches_torch_adam():
+    n = 10_000
+    p0 = torch.randn(n)
+    g = torch.randn(n)
+    # torch reference
+    p_ref = p0.clone().requires_grad_(True)
+    opt = torch.optim.Adam([p_ref], lr=1e-3)
+    p_ref.grad = g.clone()
+    for _ in range(3):
+        opt.step()
+    # fused
+    p = p0.clone().to(DEV)
+    m = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    for t in range(1, 4):
+        ops.ext().fused_adam(p, g.to(DEV), m, v, 1e-3, 0.9, 0.999, 1e-8, t)
+    assert torch.allclose(p.cpu(), p_ref.detach(), atol=1e-5)
+
+
+def test_enet_solver_matches_reference():
+    from smartcal_amd.ops import enet as enet_ops
+    torch.manual_seed(3)
+    N = M = 20
+    for trial in range(3):
+        A = torch.randn(N, M)
+        A /= A.norm()
+        y = torch.randn(N) * 0.3
+        rho1, rho2 = 0.05, 0.01
+        x_ref, opt = enet_ops.lbfgs_solve_reference(A, y, rho1, rho2)
+
+        Ab = A.unsqueeze(0).to(DEV).contiguous()
+        yb = y.unsqueeze(0).to(DEV).contiguous()
+        rho = torch.tensor([[rho1, rho2]], device=DEV)
+        xg, Yg, Sg, nh = ops.ext().enet_lbfgs_solve(Ab, yb, rho, 20, 10, 7)
+        torch.cuda.synchronize()
+
+        def loss_of(xv):
+            e = y - A @ xv
+            return float(e.dot(e) + rho1 * xv.dot(xv)
+                         + rho2 * xv.abs().sum())
+
+        l_ref = loss_of(x_ref)
+        l_gpu = loss_of(xg[0].cpu())
+        assert abs(l_gpu - l_ref) <= 0.02 * abs(l_ref) + 1e-5, \
+            f"trial {trial}: gpu loss {l_gpu} vs ref {l_ref}"
+        assert int(nh[0]) > 0
+
+
+def test_enet_influence_matches_reference():
+    from smartcal_amd.ops import enet as enet_ops
+    torch.manual_seed(4)
+    N = M = 20
+    A = torch.randn(N, M)
+    A /= A.norm()
+    y = torch.randn(N) * 0.3
+
+    Ab = A.unsqueeze(0).to(DEV).contiguous()
+    yb = y.unsqueeze(0).to(DEV).contiguous()
+    rho = torch.tensor([[0.05, 0.01]], device=DEV)
+    xg, Yg, Sg, nh = ops.ext().enet_lbfgs_solve(Ab, yb, rho, 20, 10, 7)
+    pen = torch.zeros(1, device=DEV)
+    EE, reward = ops.ext().enet_influence(Ab, yb, xg, Yg, Sg, nh, pen)
+    torch.cuda.synchronize()
+
+    # CPU oracle on the SAME curvature pairs (from the GPU solve)
+    k = int(nh[0])
+    Yc = Yg[0, :k].cpu()
+    Sc = Sg[0, :k].cpu()
+    EE_ref = enet_ops.influence_eigs_reference(A, Yc, Sc)
+    assert torch.allclose(EE[0].cpu(), EE_ref, rtol=1e-2, atol=1e-3), \
+        f"max err {(EE[0].cpu()-EE_ref).abs().max()}"
+
+    x_cpu = xg[0].cpu()
+    err = torch.norm(A @ x_cpu - y)
+    r_ref = torch.norm(y) / err + EE_ref.min() / EE_ref.max()
+    assert abs(float(reward[0]) - float(r_ref)) < 2e-2 * abs(float(r_ref)) + 1e-3
+
+
+def test_env_gpu_step_end_to_end():
+    from smartcal_amd.envs.enet import ENetEnv
+    np.random.seed(0)
+    torch.manual_seed(0)
+    env = ENetEnv(20, 20, device=DEV)
+    obs = env.reset()
+    obs2, r, done, info = env.step(np.zeros(2, dtype=np.float32))
+    assert obs2["eig"].is_cuda
+    assert torch.isfinite(r)
+    assert torch.isfinite(obs2["eig"]).all()
+
+
+def test_sac_agent_gpu_learn():
+    from smartcal_amd.rl.sac import Agent
+    np.random.seed(0)
+    torch.manual_seed(0)
+    agent = Agent(gamma=0.99, batch_size=8, n_actions=2, tau=0.005,
+                  max_mem_size=64, input_dims=[24], lr_a=1e-3, lr_c=1e-3,
+                  reward_scale=2, alpha=0.03, device=DEV)
+    obs = {"eig": torch.randn(8), "A": torch.randn(16)}
+    for _ in range(10):
+        a = agent.choose_action(obs)
+        obs2 = {"eig": torch.randn(8), "A": torch.randn(16)}
+        agent.store_transition(obs, a, 0.5, obs2, False,
+                               np.zeros(2, np.float32))
+        agent.learn()
+        obs = obs2
+    torch.cuda.synchronize()
+    assert agent.learn_counter >= 1
+    assert torch.isfinite(agent.actor_fp.flat).all()
