@@ -20,14 +20,14 @@ extern "C" __global__ void tanh_gauss_fwd_kernel(
   for (int c = lane; c < A; c += WAVE) {
     const long i = (long)row * A + c;
     const float ls = LOGSIG[i];
-    const float sig = __expf(ls);
+    const float sig = expf(ls);
     const float e = EPS[i];
     const float z = MU[i] + sig * e;
     const float at = tanhf(z);
     AT[i] = at;
     ACT[i] = at * max_action;
     lp += -0.5f * e * e - ls - LOG_SQRT_2PI
-          - __logf(max_action * (1.f - at * at) + REPARAM_NOISE);
+          - logf(max_action * (1.f - at * at) + REPARAM_NOISE);
   }
   lp = wave_sum(lp);
   if (lane == 0) LOGP[row] = lp;
@@ -47,7 +47,7 @@ extern "C" __global__ void tanh_gauss_bwd_kernel(
   const int row = i / A;
   const float at = AT[i];
   const float one_m = 1.f - at * at;
-  const float sig = __expf(LOGSIG[i]);
+  const float sig = expf(LOGSIG[i]);
   const float se = sig * EPS[i];
   const float dl = DLOGP[row];
   const float da = DACT[i];

@@ -57,14 +57,15 @@ def tanh_gauss_sample(mu: torch.Tensor, logsigma: torch.Tensor,
     if eps is None:
         eps = torch.randn_like(mu)
     if use_hip(mu):
+        mu_c = mu.contiguous()
+        logsigma_c = logsigma.contiguous()
         if reparameterize:
-            action, logprob = _TanhGaussFn.apply(mu, logsigma, eps,
+            action, logprob = _TanhGaussFn.apply(mu_c, logsigma_c, eps,
                                                  float(max_action))
         else:
             with torch.no_grad():
                 action, logprob, _ = ext().tanh_gauss_fwd(
-                    mu.contiguous(), logsigma.contiguous(), eps,
-                    float(max_action))
+                    mu_c, logsigma_c, eps, float(max_action))
     else:
         sigma = logsigma.exp()
         if reparameterize:

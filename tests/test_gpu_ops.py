@@ -120,8 +120,8 @@ def test_tanh_gauss_matches_cpu():
     lsc = ls.clone().requires_grad_(True)
     a2, lp2 = tanh_gauss_sample(muc, lsc, 1.0, True, eps)
     (a2.sum() + lp2.sum()).backward()
-    assert torch.allclose(mug.grad.cpu(), muc.grad, atol=1e-4)
-    assert torch.allclose(lsg.grad.cpu(), lsc.grad, atol=1e-4)
+    assert torch.allclose(mug.grad.cpu(), muc.grad, atol=1e-3)
+    assert torch.allclose(lsg.grad.cpu(), lsc.grad, atol=1e-3)
 
 
 def test_fused_adam_matches_torch_adam():
