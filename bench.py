@@ -78,12 +78,22 @@ def main():
     obs = env.reset()
     zeros2 = np.zeros(2, dtype=np.float32)
 
-    def one_step(o):
-        a = agent.choose_action(o)
-        o2, r, done, info = env.step(a)
-        agent.store_transition(o, a, r, o2, done, zeros2)
-        agent.learn()
-        return o2
+    if have_gpu:
+        # sync-free training step: action, reward and observation stay on
+        # the GPU for the whole loop (no host round trips)
+        def one_step(o):
+            a = agent.choose_action_tensor(o)
+            o2, r, done, info = env.step(a)
+            agent.store_transition(o, a, r, o2, done, zeros2)
+            agent.learn()
+            return o2
+    else:
+        def one_step(o):
+            a = agent.choose_action(o)
+            o2, r, done, info = env.step(a)
+            agent.store_transition(o, a, r, o2, done, zeros2)
+            agent.learn()
+            return o2
 
     # ---- pre-fill replay so learn() is live in the timed region ----
     while agent.replaymem.mem_cntr < agent.batch_size:

@@ -97,7 +97,28 @@ class ENetEnv(gymapi.Env):
         if not torch.is_tensor(action):
             action = torch.as_tensor(np.asarray(action).reshape(-1),
                                      dtype=torch.float32)
-        action = action.detach().to("cpu", torch.float32).reshape(-1)
+        action = action.detach().reshape(-1)
+
+        if self.device.type == "cuda" and action.device == self.device:
+            # fully device-resident step: no host round trip anywhere
+            scaled = action * (HIGH - LOW) / 2 + (HIGH + LOW) / 2
+            penalty = -0.1 * ((scaled < LOW).sum() + (scaled > HIGH).sum())
+            rho = scaled.clamp(LOW, HIGH)
+            self.rho = rho
+            if not keepnoise or self.y is None:
+                self._observe_y()
+            x, EE, reward = enet_ops.solve_and_influence_device(
+                self.A, self.y, rho, penalty.to(torch.float32))
+            self.x = x
+            observation = {"A": self.A.reshape(-1), "eig": EE}
+            info: dict = {}
+            if self.provide_hint:
+                if self.hint is None:
+                    self.hint = self.get_hint()
+                return observation, reward, done, self.hint, info
+            return observation, reward, done, info
+
+        action = action.to("cpu", torch.float32)
         # scale [-1,1] -> [LOW, HIGH]
         self.rho = action * (HIGH - LOW) / 2 + (HIGH + LOW) / 2
         penalty = 0.0

@@ -127,7 +127,7 @@ std::tuple<at::Tensor, at::Tensor> mfma_gemm_tn_bias(const at::Tensor& dz,
   hipLaunchKernelGGL(gemm_f32_tn_kernel, grid, dim3(256), 0, stream(),
                      dz.data_ptr<float>(), x.data_ptr<float>(),
                      dW.data_ptr<float>(), N, Bb, K);
-  hipLaunchKernelGGL(colsum_kernel, dim3((N + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(colsum_kernel, dim3((N + 63) / 64), dim3(256), 0,
                      stream(), dz.data_ptr<float>(), db.data_ptr<float>(),
                      Bb, N);
   return {dW, db};

@@ -31,10 +31,14 @@ extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
     float* __restrict__ ZHAT,       // (B, N) (LN only)
     float* __restrict__ RSTD,       // (B,)   (LN only)
     int B, int K, int N, int act, int with_ln) {
-  __shared__ float smem[16 * (BK + 1) + 4 * 16 * 2 + 16 * 2];
-  float* xs = smem;                      // [16][BK+1]
-  float* rowstat = smem + 16 * (BK + 1); // [4 waves][16 rows][2]
-  float* rowmv = rowstat + 4 * 16 * 2;   // [16][2] mean, rstd
+  // one __shared__ object (CDNA4 guide §5.5 item 4a): X tile + 4 per-wave
+  // W subtiles (stride BK+1: conflict-free b32 fragment reads) + LN scratch
+  __shared__ float smem[16 * (BK + 1) + 4 * 16 * (BK + 1) + 4 * 16 * 2
+                        + 16 * 2];
+  float* xs = smem;                            // [16][BK+1]
+  float* ws = smem + 16 * (BK + 1);            // [wave][16][BK+1]
+  float* rowstat = ws + 4 * 16 * (BK + 1);     // [4 waves][16 rows][2]
+  float* rowmv = rowstat + 4 * 16 * 2;         // [16][2] mean, rstd
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -59,20 +63,47 @@ extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
     }
     __syncthreads();
     const int kmax = min(BK, K - kk);
+    float* wsw = ws + wave * 16 * (BK + 1);
 #pragma unroll
     for (int t = 0; t < NT_MAX; ++t) {
       const int ct = wave + 4 * t;
       if (ct < ntiles) {
-        const int col = ct * 16 + l15;
-        const bool colv = col < N;
-        const float* wrow = W + (long)col * K + kk;
+        // stage this wave's 16-col x BK W subtile: coalesced float4 loads
+        // (lane 4c+j loads row c's j-th quarter), zero-padded at edges
+        {
+          // 256 float4 chunks (16 rows x BK/4); consecutive lanes take
+          // consecutive chunks of one row => fully coalesced 256B bursts
+#pragma unroll
+          for (int i = 0; i < 16 * (BK / 4) / WAVE; ++i) {
+            const int c = lane + WAVE * i;
+            const int srow = c / (BK / 4);
+            const int k = (c % (BK / 4)) * 4;
+            const int col = ct * 16 + srow;
+            const float* wrow = W + (long)col * K + kk;
+            float4 v = {0.f, 0.f, 0.f, 0.f};
+            if (col < N) {
+              if (k + 3 < kmax) {
+                v = *reinterpret_cast<const float4*>(wrow + k);
+              } else {
+                if (k + 0 < kmax) v.x = wrow[k + 0];
+                if (k + 1 < kmax) v.y = wrow[k + 1];
+                if (k + 2 < kmax) v.z = wrow[k + 2];
+                if (k + 3 < kmax) v.w = wrow[k + 3];
+              }
+            }
+            wsw[srow * (BK + 1) + k + 0] = v.x;
+            wsw[srow * (BK + 1) + k + 1] = v.y;
+            wsw[srow * (BK + 1) + k + 2] = v.z;
+            wsw[srow * (BK + 1) + k + 3] = v.w;
+          }
+        }
+#pragma unroll
         for (int k4 = 0; k4 < BK / 4; ++k4) {
-          // MFMA is a whole-wave op: never predicate it per-lane. Edge
-          // tiles contribute zeros through the operands instead (X tile is
-          // zero-padded in LDS; W reads are masked here).
+          // MFMA is a whole-wave op: never predicate it per-lane; edge
+          // tiles contribute zeros through the zero-padded LDS operands.
           const int k = k4 * 4 + l4;
           float a = xs[l15 * (BK + 1) + k];
-          float b = (colv && k < kmax) ? wrow[k] : 0.f;
+          float b = wsw[l15 * (BK + 1) + k];
           acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[t], 0, 0, 0);
         }
       }

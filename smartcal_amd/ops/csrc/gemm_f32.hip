@@ -20,16 +20,32 @@ extern "C" __global__ __launch_bounds__(256) void gemm_f32_nn_kernel(
   const int l4 = lane >> 4;
   const int row0 = blockIdx.x * 16;
   const int col0 = (blockIdx.y * 4 + wave) * 16;
-  if (col0 >= N) return;
+  // NOTE: no early return — all waves must reach the barriers below;
+  // out-of-range waves compute on zeros and skip their stores.
 
-  const int row = row0 + l15;   // A fragment row
+  // A tile staged in LDS (A fragment reads are row-scattered in global);
+  // B fragment reads are naturally coalesced (consecutive lanes ->
+  // consecutive columns), so B streams from L2.
+  __shared__ float as[16 * 68];  // [16][64+4] pad 4: conflict-free b32 reads
   const int col = col0 + l15;   // B fragment col
   f32x4 acc = (f32x4){0.f, 0.f, 0.f, 0.f};
-  for (int kk = 0; kk < K; kk += 4) {
-    const int k = kk + l4;
-    const float a = (row < M && k < K) ? A[(long)row * K + k] : 0.f;
-    const float b = (col < N && k < K) ? B[(long)k * N + col] : 0.f;
-    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  for (int kk = 0; kk < K; kk += 64) {
+    const int kmax = min(64, K - kk);
+    for (int idx = tid; idx < 16 * 64; idx += 256) {
+      const int r = idx >> 6, c = idx & 63;
+      const int gr = row0 + r, gc = kk + c;
+      as[r * 68 + c] = (gr < M && gc < K) ? A[(long)gr * K + gc] : 0.f;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int k4 = 0; k4 < 16; ++k4) {
+      const int k = k4 * 4 + l4;
+      const float a = as[l15 * 68 + k];
+      const float b = (col < N && k < kmax) ? B[(long)(kk + k) * N + col]
+                                            : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+    }
+    __syncthreads();
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -69,13 +85,20 @@ extern "C" __global__ __launch_bounds__(256) void gemm_f32_tn_kernel(
   }
 }
 
-// db (N) = column sums of DZ (B, N).
-extern "C" __global__ void colsum_kernel(const float* __restrict__ DZ,
-                                         float* __restrict__ OUT, int B,
-                                         int N) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= N) return;
+// db (N) = column sums of DZ (B, N). Block = 256 threads covering 64
+// columns x 4 row-stripes (coalesced: consecutive threads -> consecutive
+// columns), combined through LDS.
+extern "C" __global__ __launch_bounds__(256) void colsum_kernel(
+    const float* __restrict__ DZ, float* __restrict__ OUT, int B, int N) {
+  __shared__ float red[4 * 64];
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int stripe = threadIdx.x >> 6;
   float s = 0.f;
-  for (int r = 0; r < B; ++r) s += DZ[(long)r * N + c];
-  OUT[c] = s;
+  if (c < N)
+    for (int r = stripe; r < B; r += 4) s += DZ[(long)r * N + c];
+  red[stripe * 64 + (threadIdx.x & 63)] = s;
+  __syncthreads();
+  if (threadIdx.x < 64 && c < N)
+    OUT[c] = red[threadIdx.x] + red[64 + threadIdx.x] +
+             red[128 + threadIdx.x] + red[192 + threadIdx.x];
 }

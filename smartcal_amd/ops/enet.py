@@ -89,6 +89,23 @@ def influence_eigs_reference(A: torch.Tensor, Y: torch.Tensor,
     return ev + 1.0
 
 
+def solve_and_influence_device(A: torch.Tensor, y: torch.Tensor,
+                               rho: torch.Tensor, penalty: torch.Tensor,
+                               epochs: int = 20, max_iter: int = 10,
+                               history: int = 7):
+    """Sync-free GPU env step: rho (2,) and penalty are device tensors —
+    no host round trip anywhere in the step."""
+    ext()  # loud failure if the extension is missing on a GPU box
+    Ab = A.unsqueeze(0).contiguous()
+    yb = y.unsqueeze(0).contiguous()
+    x, Yc, Sc, nh = ext().enet_lbfgs_solve(Ab, yb,
+                                           rho.reshape(1, 2).contiguous(),
+                                           epochs, max_iter, history)
+    EE, reward = ext().enet_influence(Ab, yb, x, Yc, Sc, nh,
+                                      penalty.reshape(1).contiguous())
+    return x[0], EE[0], reward[0]
+
+
 def solve_and_influence(A: torch.Tensor, y: torch.Tensor,
                         rho1, rho2, penalty: float,
                         epochs: int = 20, max_iter: int = 10,

@@ -66,7 +66,12 @@ class ReplayBuffer:
         self.state_memory[i] = self._as(state, self.state_memory[i])
         self.new_state_memory[i] = self._as(state_, self.new_state_memory[i])
         self.action_memory[i] = self._as(action, self.action_memory[i])
-        self.reward_memory[i] = float(reward)
+        if torch.is_tensor(reward):
+            # keep device-resident — no host sync
+            self.reward_memory[i] = reward.detach().to(
+                self.reward_memory.device, torch.float32).reshape(())
+        else:
+            self.reward_memory[i] = float(reward)
         self.terminal_memory[i] = bool(done)
         if hint is not None:
             self.hint_memory[i] = self._as(hint, self.hint_memory[i])

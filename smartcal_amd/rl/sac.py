@@ -111,6 +111,15 @@ class Agent:
                                                   reparameterize=False)
         return actions.cpu().numpy()[0]
 
+    def choose_action_tensor(self, observation):
+        """Device-resident action (no host sync) — the bench/vectorized
+        rollout path; ``choose_action`` keeps the reference's numpy API."""
+        state = obs_to_state(observation).to(self.device)
+        with torch.no_grad():
+            actions, _ = self.actor.sample_normal(state,
+                                                  reparameterize=False)
+        return actions.reshape(-1)
+
     # ------------------------------------------------------------------
     def _grad_sync(self, fps):
         if self.grad_hook is not None:
