@@ -62,7 +62,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_fwd(
   check_f32(W, "W");
   const int B = x.size(0), K = x.size(1), N = W.size(0);
   TORCH_CHECK(W.size(1) == K, "W/K mismatch");
-  TORCH_CHECK(N <= 576, "fused_linear: N>576 unsupported (extend NT_MAX)");
+  TORCH_CHECK(N <= 640, "fused_linear: N>640 unsupported (extend NT_MAX)");
   TORCH_CHECK(!with_ln || gamma.has_value(), "LN requires gamma");
   auto y = at::empty({B, N}, x.options());
   auto zhat = with_ln ? at::empty({B, N}, x.options())
@@ -70,7 +70,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_fwd(
   auto rstd = with_ln ? at::empty({B}, x.options())
                       : at::empty({0}, x.options());
   dim3 grid((B + 15) / 16);
-  hipLaunchKernelGGL(fused_linear_fwd_kernel, grid, dim3(256), 0, stream(),
+  hipLaunchKernelGGL(fused_linear_fwd_kernel, grid, dim3(512), 0, stream(),
                      x.data_ptr<float>(), W.data_ptr<float>(),
                      bias ? bias->data_ptr<float>() : nullptr,
                      gamma ? gamma->data_ptr<float>() : nullptr,

@@ -2,15 +2,18 @@
 // actor/critic MLP hot op (reference composes
 // Linear -> LayerNorm -> ELU as separate torch ops, enet_sac.py:436-444).
 //
-// Forward kernel: one workgroup (4 waves, 256 threads) owns a 16-row slab of
+// Forward kernel: one workgroup (8 waves, 512 threads) owns a 16-row slab of
 // the batch and ALL output columns (so the LayerNorm row statistics never
 // leave the workgroup). The GEMM runs on v_mfma_f32_16x16x4_f32 (exact f32,
-// §3 of the CDNA4 guide) with the X tile staged in LDS (+1 pad, conflict
-// free) and W fragments streamed from L2 (W is re-read only gridDim.x =
-// ceil(B/16) times; RL batches are 1..256 so W stays L2-resident). The
-// epilogue fuses bias + LayerNorm (shfl_xor 16-lane row reductions + one
-// LDS cross-wave combine) + activation, and stores y, zhat, rstd for the
-// backward pass. No separate normalization kernel, no extra HBM round trip.
+// §3 of the CDNA4 guide; no TF32 on gfx950) with the X tile staged in LDS
+// (+1 pad, conflict free) and each wave register-prefetching its NEXT
+// 16-col W subtile (coalesced float4 bursts) while the current one runs on
+// the matrix cores — the T14 issue-early/write-late split of the CDNA4
+// guide, sized for RL shapes where W is L2-resident and the kernel is
+// latency-bound. The epilogue fuses bias + LayerNorm (shfl_xor 16-lane row
+// reductions + one LDS cross-wave combine) + activation, storing y, zhat,
+// rstd for backward. No separate normalization kernel, no extra HBM round
+// trip.
 //
 // Backward: (1) ln_act_bwd_kernel — act'(y) + LayerNorm backward (row
 // reductions in-workgroup) + dgamma/dbeta column partials; (2)/(3) are MFMA
@@ -18,10 +21,12 @@
 
 #include "common.h"
 
-#define NT_MAX 9           // max 16-col tiles per wave => N <= 4*NT_MAX*16
+#define NWAVES 8
+#define NT_MAX 5           // max 16-col tiles per wave => N <= 16*NWAVES*NT_MAX = 640
 #define BK 64              // K-tile staged in LDS
+#define WCH (16 * (BK / 4) / WAVE)  // float4 chunks per lane per W subtile (=4)
 
-extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
+extern "C" __global__ __launch_bounds__(512) void fused_linear_fwd_kernel(
     const float* __restrict__ X,    // (B, K)
     const float* __restrict__ W,    // (N, K) row-major
     const float* __restrict__ bias, // (N) or null
@@ -31,14 +36,14 @@ extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
     float* __restrict__ ZHAT,       // (B, N) (LN only)
     float* __restrict__ RSTD,       // (B,)   (LN only)
     int B, int K, int N, int act, int with_ln) {
-  // one __shared__ object (CDNA4 guide §5.5 item 4a): X tile + 4 per-wave
-  // W subtiles (stride BK+1: conflict-free b32 fragment reads) + LN scratch
-  __shared__ float smem[16 * (BK + 1) + 4 * 16 * (BK + 1) + 4 * 16 * 2
-                        + 16 * 2];
-  float* xs = smem;                            // [16][BK+1]
-  float* ws = smem + 16 * (BK + 1);            // [wave][16][BK+1]
-  float* rowstat = ws + 4 * 16 * (BK + 1);     // [4 waves][16 rows][2]
-  float* rowmv = rowstat + 4 * 16 * 2;         // [16][2] mean, rstd
+  // one __shared__ object (CDNA4 guide §5.5 item 4a): X tile + NWAVES
+  // per-wave W subtiles (stride BK+1: conflict-free b32 fragment reads)
+  __shared__ float smem[16 * (BK + 1) + NWAVES * 16 * (BK + 1)
+                        + NWAVES * 16 * 2 + 16 * 2];
+  float* xs = smem;                               // [16][BK+1]
+  float* ws = smem + 16 * (BK + 1);               // [wave][16][BK+1]
+  float* rowstat = ws + NWAVES * 16 * (BK + 1);   // [NWAVES][16 rows][2]
+  float* rowmv = rowstat + NWAVES * 16 * 2;       // [16][2] mean, rstd
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -49,13 +54,25 @@ extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
 
   const int ntiles = (N + 15) >> 4;  // 16-col tiles over N
 
+  // per-lane W-chunk coordinates (fixed across tiles)
+  const int wc_row[WCH] = {(lane + 0 * WAVE) / (BK / 4),
+                           (lane + 1 * WAVE) / (BK / 4),
+                           (lane + 2 * WAVE) / (BK / 4),
+                           (lane + 3 * WAVE) / (BK / 4)};
+  const int wc_k[WCH] = {((lane + 0 * WAVE) % (BK / 4)) * 4,
+                         ((lane + 1 * WAVE) % (BK / 4)) * 4,
+                         ((lane + 2 * WAVE) % (BK / 4)) * 4,
+                         ((lane + 3 * WAVE) % (BK / 4)) * 4};
+
   f32x4 acc[NT_MAX];
 #pragma unroll
   for (int t = 0; t < NT_MAX; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
+  float* wsw = ws + wave * 16 * (BK + 1);
+
   for (int kk = 0; kk < K; kk += BK) {
     // stage X[row0..row0+15][kk..kk+BK) into LDS (zero-padded)
-    for (int idx = tid; idx < 16 * BK; idx += 256) {
+    for (int idx = tid; idx < 16 * BK; idx += 512) {
       int r = idx / BK, c = idx % BK;
       int gr = row0 + r, gc = kk + c;
       xs[r * (BK + 1) + c] =
@@ -63,61 +80,76 @@ extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
     }
     __syncthreads();
     const int kmax = min(BK, K - kk);
-    float* wsw = ws + wave * 16 * (BK + 1);
+
+    // coalesced, guarded float4 load of one W subtile into registers
+    auto load_wtile = [&](int t, float4* vr) {
 #pragma unroll
-    for (int t = 0; t < NT_MAX; ++t) {
-      const int ct = wave + 4 * t;
-      if (ct < ntiles) {
-        // stage this wave's 16-col x BK W subtile: coalesced float4 loads
-        // (lane 4c+j loads row c's j-th quarter), zero-padded at edges
-        {
-          // 256 float4 chunks (16 rows x BK/4); consecutive lanes take
-          // consecutive chunks of one row => fully coalesced 256B bursts
-#pragma unroll
-          for (int i = 0; i < 16 * (BK / 4) / WAVE; ++i) {
-            const int c = lane + WAVE * i;
-            const int srow = c / (BK / 4);
-            const int k = (c % (BK / 4)) * 4;
-            const int col = ct * 16 + srow;
-            const float* wrow = W + (long)col * K + kk;
-            float4 v = {0.f, 0.f, 0.f, 0.f};
-            if (col < N) {
-              if (k + 3 < kmax) {
-                v = *reinterpret_cast<const float4*>(wrow + k);
-              } else {
-                if (k + 0 < kmax) v.x = wrow[k + 0];
-                if (k + 1 < kmax) v.y = wrow[k + 1];
-                if (k + 2 < kmax) v.z = wrow[k + 2];
-                if (k + 3 < kmax) v.w = wrow[k + 3];
-              }
-            }
-            wsw[srow * (BK + 1) + k + 0] = v.x;
-            wsw[srow * (BK + 1) + k + 1] = v.y;
-            wsw[srow * (BK + 1) + k + 2] = v.z;
-            wsw[srow * (BK + 1) + k + 3] = v.w;
+      for (int i = 0; i < WCH; ++i) {
+        const int ct = wave + NWAVES * t;
+        const int col = ct * 16 + wc_row[i];
+        const int k = wc_k[i];
+        float4 v = {0.f, 0.f, 0.f, 0.f};
+        if (ct < ntiles && col < N) {
+          const float* wrow = W + (long)col * K + kk;
+          if (k + 3 < kmax) {
+            v = *reinterpret_cast<const float4*>(wrow + k);
+          } else {
+            if (k + 0 < kmax) v.x = wrow[k + 0];
+            if (k + 1 < kmax) v.y = wrow[k + 1];
+            if (k + 2 < kmax) v.z = wrow[k + 2];
+            if (k + 3 < kmax) v.w = wrow[k + 3];
           }
         }
+        vr[i] = v;
+      }
+    };
+    auto write_wtile = [&](const float4* vr) {
 #pragma unroll
-        for (int k4 = 0; k4 < BK / 4; ++k4) {
-          // MFMA is a whole-wave op: never predicate it per-lane; edge
-          // tiles contribute zeros through the zero-padded LDS operands.
-          const int k = k4 * 4 + l4;
-          float a = xs[l15 * (BK + 1) + k];
-          float b = wsw[l15 * (BK + 1) + k];
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[t], 0, 0, 0);
-        }
+      for (int i = 0; i < WCH; ++i) {
+        const int o = wc_row[i] * (BK + 1) + wc_k[i];
+        wsw[o + 0] = vr[i].x;
+        wsw[o + 1] = vr[i].y;
+        wsw[o + 2] = vr[i].z;
+        wsw[o + 3] = vr[i].w;
+      }
+    };
+
+    // software pipeline over this wave's column tiles: write tile t to
+    // LDS, issue tile t+1's global loads, run tile t's MFMAs (the global
+    // latency hides under the MFMA + ds traffic — T14)
+    float4 wa[WCH], wb[WCH];
+    load_wtile(0, wa);
+#pragma unroll
+    for (int t = 0; t < NT_MAX; ++t) {
+      const int ct = wave + NWAVES * t;
+      if (ct >= ntiles) break;
+      if (t % 2 == 0) {
+        write_wtile(wa);
+        if (t + 1 < NT_MAX) load_wtile(t + 1, wb);
+      } else {
+        write_wtile(wb);
+        if (t + 1 < NT_MAX) load_wtile(t + 1, wa);
+      }
+#pragma unroll
+      for (int k4 = 0; k4 < BK / 4; ++k4) {
+        // MFMA is a whole-wave op: never predicate it per-lane; edge
+        // tiles contribute zeros through the zero-padded LDS operands.
+        const int k = k4 * 4 + l4;
+        float a = xs[l15 * (BK + 1) + k];
+        float b = wsw[l15 * (BK + 1) + k];
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[t], 0, 0, 0);
       }
     }
     __syncthreads();
   }
 
   // ---- epilogue: bias + LN + act ----
-  // acc[t][r] holds z[row = row0 + l4*4 + r][col = (wave+4t)*16 + l15]
+  // acc[t][r] holds z[row = row0 + l4*4 + r][col = (wave+NWAVES*t)*16 + l15]
   float zrow[NT_MAX][4];
   float psum[4] = {0.f, 0.f, 0.f, 0.f}, psq[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
   for (int t = 0; t < NT_MAX; ++t) {
-    const int ct = wave + 4 * t;
+    const int ct = wave + NWAVES * t;
     const int col = ct * 16 + l15;
     const bool colv = (ct < ntiles) && (col < N);
     const float bv = (colv && bias) ? bias[col] : 0.f;
@@ -146,13 +178,13 @@ extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
     }
     __syncthreads();
     if (tid < 16) {
-      float s = 0.f, q = 0.f;
+      float sa = 0.f, q = 0.f;
 #pragma unroll
-      for (int w = 0; w < 4; ++w) {
-        s += rowstat[(w * 16 + tid) * 2 + 0];
+      for (int w = 0; w < NWAVES; ++w) {
+        sa += rowstat[(w * 16 + tid) * 2 + 0];
         q += rowstat[(w * 16 + tid) * 2 + 1];
       }
-      const float mean = s / N;
+      const float mean = sa / N;
       float var = q / N - mean * mean;
       const float rstd = rsqrtf(fmaxf(var, 0.f) + 1e-5f);
       rowmv[tid * 2 + 0] = mean;
@@ -175,7 +207,7 @@ extern "C" __global__ __launch_bounds__(256) void fused_linear_fwd_kernel(
     }
 #pragma unroll
     for (int t = 0; t < NT_MAX; ++t) {
-      const int ct = wave + 4 * t;
+      const int ct = wave + NWAVES * t;
       const int col = ct * 16 + l15;
       if (ct < ntiles && col < N) {
         float z = zrow[t][r];
