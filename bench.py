@@ -99,6 +99,16 @@ def main():
     while agent.replaymem.mem_cntr < agent.batch_size:
         obs = one_step(obs)
 
+    if have_gpu:
+        # capture the whole learn step into one hipGraph (forwards,
+        # backwards, fused Adam, polyak — single replay per step)
+        try:
+            agent.enable_cuda_graph()
+        except Exception as e:  # noqa: BLE001
+            import sys as _sys
+            print(f"[bench] graph capture unavailable, eager path: {e}",
+                  file=_sys.stderr)
+
     # ---- warmup ----
     for _ in range(args.warmup):
         obs = one_step(obs)

@@ -32,8 +32,9 @@ extern "C" __global__ void tanh_gauss_bwd_kernel(const float*, const float*,
                                                  const float*, float*, float*,
                                                  float, int, int);
 extern "C" __global__ void fused_adam_kernel(float*, const float*, float*,
-                                             float*, float, float, float,
-                                             float, float, float, long);
+                                             float*, float*, float, float,
+                                             float, float, long);
+extern "C" __global__ void adam_bump_kernel(float*);
 extern "C" __global__ void enet_lbfgs_solve_kernel(
     const float*, const float*, const float*, float*, float*, float*, int*,
     int, int, int, int, int, int);
@@ -168,16 +169,17 @@ std::tuple<at::Tensor, at::Tensor> tanh_gauss_bwd(
 }
 
 void fused_adam(at::Tensor& p, const at::Tensor& g, at::Tensor& m,
-                at::Tensor& v, double lr, double b1, double b2, double eps,
-                int64_t t) {
+                at::Tensor& v, at::Tensor& t_dev, double lr, double b1,
+                double b2, double eps) {
   check_f32(p, "p");
   const long n = p.numel();
-  const float bc1 = 1.f - powf((float)b1, (float)t);
-  const float bc2 = 1.f - powf((float)b2, (float)t);
   hipLaunchKernelGGL(fused_adam_kernel, dim3((n + 255) / 256), dim3(256), 0,
                      stream(), p.data_ptr<float>(), g.data_ptr<float>(),
-                     m.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
-                     (float)b1, (float)b2, (float)eps, bc1, bc2, n);
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     t_dev.data_ptr<float>(), (float)lr, (float)b1,
+                     (float)b2, (float)eps, n);
+  hipLaunchKernelGGL(adam_bump_kernel, dim3(1), dim3(64), 0, stream(),
+                     t_dev.data_ptr<float>());
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> enet_lbfgs_solve(

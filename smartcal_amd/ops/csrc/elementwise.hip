@@ -59,12 +59,18 @@ extern "C" __global__ void tanh_gauss_bwd_kernel(
 }
 
 // Fused Adam over a flat parameter pool (bias-corrected, torch defaults).
+// The step counter lives on the DEVICE (tptr) so the whole optimizer step
+// is hipGraph-capturable: block 0 thread 0 bumps it, every block computes
+// the bias corrections from the pre-bump value.
 extern "C" __global__ void fused_adam_kernel(
     float* __restrict__ P, const float* __restrict__ G,
-    float* __restrict__ M, float* __restrict__ V, float lr, float b1,
-    float b2, float eps, float bc1, float bc2, long n) {
+    float* __restrict__ M, float* __restrict__ V, float* __restrict__ tptr,
+    float lr, float b1, float b2, float eps, long n) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const float t = tptr[0] + 1.f;
   if (i >= n) return;
+  const float bc1 = 1.f - powf(b1, t);
+  const float bc2 = 1.f - powf(b2, t);
   const float g = G[i];
   const float m = b1 * M[i] + (1.f - b1) * g;
   const float v = b2 * V[i] + (1.f - b2) * g * g;
@@ -73,4 +79,9 @@ extern "C" __global__ void fused_adam_kernel(
   const float mhat = m / bc1;
   const float vhat = v / bc2;
   P[i] -= lr * mhat / (sqrtf(vhat) + eps);
+}
+
+// separate tiny bump kernel so the ordering vs fused_adam_kernel is safe
+extern "C" __global__ void adam_bump_kernel(float* __restrict__ tptr) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) tptr[0] += 1.f;
 }

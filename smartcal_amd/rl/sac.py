@@ -125,25 +125,10 @@ class Agent:
         if self.grad_hook is not None:
             self.grad_hook(fps)
 
-    def learn(self):
-        if self.replaymem.mem_cntr < self.batch_size:
-            return
-
-        if self.prioritized:
-            (state, action, reward, new_state, done, hint), idxs, is_w = \
-                self.replaymem.sample_buffer(self.batch_size)
-            is_w = is_w.to(self.device).unsqueeze(1)
-        else:
-            state, action, reward, new_state, done, hint = \
-                self.replaymem.sample_buffer(self.batch_size)
-
-        state_batch = state.to(self.device)
-        new_state_batch = new_state.to(self.device)
-        action_batch = action.to(self.device)
-        reward_batch = self.scale * reward.to(self.device).unsqueeze(1)
-        terminal_batch = done.to(self.device).unsqueeze(1)
-        hint_batch = hint.to(self.device)
-
+    def _learn_body(self, state_batch, new_state_batch, action_batch,
+                    reward_batch, terminal_batch, hint_batch, is_w=None):
+        """Tensor-only learn step (hipGraph-capturable on GPU): soft target,
+        twin-critic regression, actor update, fused Adam, polyak."""
         with torch.no_grad():
             new_actions, new_log_probs = self.actor.sample_normal(
                 new_state_batch, reparameterize=False)
@@ -151,16 +136,17 @@ class Agent:
             q2_t = self.target_critic_2(new_state_batch, new_actions)
             min_next_target = torch.min(q1_t, q2_t) \
                 - self.alpha * new_log_probs
-            min_next_target[terminal_batch] = 0.0
+            # masked_fill (not boolean indexing): same semantics as the
+            # reference's t[mask]=0 but static-shaped => graph-safe
+            min_next_target = min_next_target.masked_fill(terminal_batch,
+                                                          0.0)
             new_q_value = reward_batch + self.gamma * min_next_target
 
         q1 = self.critic_1(state_batch, action_batch)
         q2 = self.critic_2(state_batch, action_batch)
-        if self.prioritized:
+        if is_w is not None:
             critic_1_loss = (is_w * (q1 - new_q_value).pow(2)).mean()
             critic_2_loss = (is_w * (q2 - new_q_value).pow(2)).mean()
-            td = 0.5 * ((q1 - new_q_value).abs() + (q2 - new_q_value).abs())
-            self.replaymem.update_priorities(idxs, td)
         else:
             critic_1_loss = F.mse_loss(q1, new_q_value)
             critic_2_loss = F.mse_loss(q2, new_q_value)
@@ -190,6 +176,120 @@ class Agent:
         actor_loss.backward()
         self._grad_sync([self.actor_fp])
         self.actor_opt.step()
+        self.update_network_parameters()
+        return q1, new_q_value
+
+    # -- hipGraph capture of the learn step -------------------------------
+    def enable_cuda_graph(self):
+        """Capture the whole learn step (forwards, backwards, fused Adam,
+        polyak) into ONE hipGraph. Per learn step afterwards: one randint,
+        four index_select gathers into static buffers, one graph replay —
+        instead of ~100 python-dispatched launches. Only for the plain
+        (non-PER, non-hint) path and after the replay has batch_size
+        entries."""
+        assert not self.use_hint and not self.prioritized
+        assert self.device.type == "cuda"
+        B = self.batch_size
+        mem = self.replaymem
+        self._g_idx = torch.zeros(B, dtype=torch.long, device=self.device)
+        self._g_state = torch.zeros(B, mem.state_memory.shape[1],
+                                    device=self.device)
+        self._g_new_state = torch.zeros_like(self._g_state)
+        self._g_action = torch.zeros(B, self.n_actions, device=self.device)
+        self._g_reward = torch.zeros(B, 1, device=self.device)
+        self._g_done = torch.zeros(B, 1, dtype=torch.bool,
+                                   device=self.device)
+        self._g_hint = torch.zeros(B, self.n_actions, device=self.device)
+
+        # populate statics with a real sample so warmup trains on data
+        n = len(mem)
+        if n > 0:
+            torch.randint(0, n, (B,), device=self.device, out=self._g_idx)
+            torch.index_select(mem.state_memory, 0, self._g_idx,
+                               out=self._g_state)
+            torch.index_select(mem.new_state_memory, 0, self._g_idx,
+                               out=self._g_new_state)
+            torch.index_select(mem.action_memory, 0, self._g_idx,
+                               out=self._g_action)
+            torch.index_select(mem.reward_memory, 0, self._g_idx,
+                               out=self._g_reward.view(-1))
+            self._g_reward.mul_(self.scale)
+            torch.index_select(mem.terminal_memory, 0, self._g_idx,
+                               out=self._g_done.view(-1))
+
+        # warmup on a side stream (allocator + autograd settle)
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self._learn_body(self._g_state, self._g_new_state,
+                                 self._g_action, self._g_reward,
+                                 self._g_done, self._g_hint)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._learn_body(self._g_state, self._g_new_state,
+                             self._g_action, self._g_reward,
+                             self._g_done, self._g_hint)
+        self._graph = g
+
+    def _learn_graphed(self):
+        mem = self.replaymem
+        n = len(mem)
+        torch.randint(0, n, (self.batch_size,), device=self.device,
+                      out=self._g_idx)
+        torch.index_select(mem.state_memory, 0, self._g_idx,
+                           out=self._g_state)
+        torch.index_select(mem.new_state_memory, 0, self._g_idx,
+                           out=self._g_new_state)
+        torch.index_select(mem.action_memory, 0, self._g_idx,
+                           out=self._g_action)
+        torch.index_select(mem.reward_memory, 0, self._g_idx,
+                           out=self._g_reward.view(-1))
+        self._g_reward.mul_(self.scale)  # body expects scaled rewards
+        torch.index_select(mem.terminal_memory, 0, self._g_idx,
+                           out=self._g_done.view(-1))
+        self._graph.replay()
+        self.learn_counter += 1
+
+    def learn(self):
+        if self.replaymem.mem_cntr < self.batch_size:
+            return
+
+        if getattr(self, "_graph", None) is not None:
+            return self._learn_graphed()
+
+        if self.prioritized:
+            (state, action, reward, new_state, done, hint), idxs, is_w = \
+                self.replaymem.sample_buffer(self.batch_size)
+            is_w = is_w.to(self.device).unsqueeze(1)
+        else:
+            state, action, reward, new_state, done, hint = \
+                self.replaymem.sample_buffer(self.batch_size)
+            is_w = None
+
+        state_batch = state.to(self.device)
+        new_state_batch = new_state.to(self.device)
+        action_batch = action.to(self.device)
+        reward_batch = self.scale * reward.to(self.device).unsqueeze(1)
+        terminal_batch = done.to(self.device).unsqueeze(1)
+        hint_batch = hint.to(self.device)
+
+        if self.prioritized:
+            q1, new_q_value = self._learn_body(
+                state_batch, new_state_batch, action_batch, reward_batch,
+                terminal_batch, hint_batch, is_w)
+            with torch.no_grad():
+                q2 = self.critic_2(state_batch, action_batch)
+                td = 0.5 * ((q1 - new_q_value).abs()
+                            + (q2 - new_q_value).abs())
+            self.replaymem.update_priorities(idxs, td)
+        else:
+            self._learn_body(state_batch, new_state_batch, action_batch,
+                             reward_batch, terminal_batch, hint_batch)
 
         if self.learn_counter % 10 == 0 and (self.learn_alpha
                                              or self.use_hint):
@@ -209,7 +309,6 @@ class Agent:
                     self.rho = self.rho + self.admm_rho * gfun
 
         self.learn_counter += 1
-        self.update_network_parameters()
 
     # -- checkpointing (reference file layout: <name>_sac_<role>.model) ---
     def _path(self, name):

@@ -85,6 +85,10 @@ class FusedAdam:
         self.step_count = 0
         self.m = torch.zeros_like(fp.flat)
         self.v = torch.zeros_like(fp.flat)
+        # device-resident step counter so the whole Adam step (incl. bias
+        # correction) is hipGraph-capturable
+        self.t_dev = torch.zeros(1, device=fp.flat.device,
+                                 dtype=torch.float32)
 
     def zero_grad(self):
         self.fp.zero_grad()
@@ -98,8 +102,8 @@ class FusedAdam:
         if self.weight_decay:
             g = g.add(self.fp.flat, alpha=self.weight_decay)
         if use_hip(self.fp.flat):
-            ext().fused_adam(self.fp.flat, g, self.m, self.v, self.lr,
-                             self.beta1, self.beta2, self.eps, t)
+            ext().fused_adam(self.fp.flat, g, self.m, self.v, self.t_dev,
+                             self.lr, self.beta1, self.beta2, self.eps)
         else:
             self.m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
             self.v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
@@ -110,10 +114,14 @@ class FusedAdam:
 
     def state_dict(self):
         return {"step": self.step_count, "m": self.m, "v": self.v,
-                "lr": self.lr}
+                "lr": self.lr, "t_dev": self.t_dev}
 
     def load_state_dict(self, sd):
         self.step_count = sd["step"]
         self.m.copy_(sd["m"])
         self.v.copy_(sd["v"])
         self.lr = sd.get("lr", self.lr)
+        if "t_dev" in sd:
+            self.t_dev.copy_(sd["t_dev"])
+        else:
+            self.t_dev.fill_(float(self.step_count))

@@ -138,8 +138,11 @@ def test_fused_adam_matches_torch_adam():
     p = p0.clone().to(DEV)
     m = torch.zeros(n, device=DEV)
     v = torch.zeros(n, device=DEV)
-    for t in range(1, 4):
-        ops.ext().fused_adam(p, g.to(DEV), m, v, 1e-3, 0.9, 0.999, 1e-8, t)
+    t_dev = torch.zeros(1, device=DEV)
+    for _ in range(3):
+        ops.ext().fused_adam(p, g.to(DEV), m, v, t_dev, 1e-3, 0.9, 0.999,
+                             1e-8)
+    assert float(t_dev) == 3.0
     assert torch.allclose(p.cpu(), p_ref.detach(), atol=1e-5)
 
 
@@ -212,6 +215,33 @@ def test_env_gpu_step_end_to_end():
     assert obs2["eig"].is_cuda
     assert torch.isfinite(r)
     assert torch.isfinite(obs2["eig"]).all()
+
+
+def test_sac_graphed_learn():
+    from smartcal_amd.rl.sac import Agent
+    np.random.seed(0)
+    torch.manual_seed(0)
+    agent = Agent(gamma=0.99, batch_size=8, n_actions=2, tau=0.005,
+                  max_mem_size=64, input_dims=[24], lr_a=1e-3, lr_c=1e-3,
+                  reward_scale=2, alpha=0.03, device=DEV)
+    obs = {"eig": torch.randn(8), "A": torch.randn(16)}
+    for _ in range(10):
+        a = agent.choose_action(obs)
+        obs2 = {"eig": torch.randn(8), "A": torch.randn(16)}
+        agent.store_transition(obs, a, 0.5, obs2, False,
+                               np.zeros(2, np.float32))
+        obs = obs2
+    agent.enable_cuda_graph()
+    before = agent.actor_fp.flat.clone()
+    for _ in range(5):
+        agent.learn()
+    torch.cuda.synchronize()
+    assert agent.learn_counter == 5
+    assert torch.isfinite(agent.actor_fp.flat).all()
+    assert not torch.allclose(before, agent.actor_fp.flat)
+    assert torch.isfinite(agent.critic_1_fp.flat).all()
+    # polyak kept targets tracking
+    assert torch.isfinite(agent.target_critic_1_fp.flat).all()
 
 
 def test_sac_agent_gpu_learn():
