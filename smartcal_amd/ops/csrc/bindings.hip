@@ -196,7 +196,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> enet_lbfgs_solve(
   auto S = at::empty({E, (long)HMAX, M}, A.options());
   auto nh = at::empty({E}, A.options().dtype(at::kInt));
   const int lds_floats = N * M + N + 5 * M + N + 2 * HMAX * M + 2 * M
-                         + 2 * HMAX;
+                         + 2 * HMAX + M;
   hipLaunchKernelGGL(enet_lbfgs_solve_kernel, dim3(E), dim3(64),
                      lds_floats * sizeof(float), stream(),
                      A.data_ptr<float>(), y.data_ptr<float>(),

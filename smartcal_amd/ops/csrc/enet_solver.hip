@@ -27,20 +27,21 @@
 #define NMAX 32
 
 struct LdsLayout {
-  float* A;      // N*M
-  float* y;      // N
-  float* x;      // M
-  float* g;      // M
-  float* gprev;  // M
-  float* d;      // M
-  float* x0;     // M
-  float* r;      // N
-  float* S;      // HMAX*M
-  float* Yh;     // HMAX*M
-  float* bg0;    // M (bracket grad low)
-  float* bg1;    // M (bracket grad high)
-  float* ro;     // HMAX (uniform scalars; LDS to avoid scratch spills)
-  float* al;     // HMAX
+  float* __restrict__ A;      // N*M
+  float* __restrict__ y;      // N
+  float* __restrict__ x;      // M
+  float* __restrict__ g;      // M
+  float* __restrict__ gprev;  // M
+  float* __restrict__ d;      // M
+  float* __restrict__ x0;     // M
+  float* __restrict__ r;      // N
+  float* __restrict__ S;      // HMAX*M
+  float* __restrict__ Yh;     // HMAX*M
+  float* __restrict__ bg0;    // M (bracket grad low)
+  float* __restrict__ bg1;    // M (bracket grad high)
+  float* __restrict__ ro;     // HMAX (uniform scalars; LDS to avoid scratch spills)
+  float* __restrict__ al;     // HMAX
+  float* __restrict__ xprev;  // M (previous-epoch x snapshot)
 };
 
 // residual + loss + gradient of ||y-Ax||^2 + rho1||x||^2 + rho2||x||_1
@@ -49,19 +50,34 @@ __device__ static float eval_loss_grad(const LdsLayout& L, int N, int M,
   const int lane = threadIdx.x;
   float ri = 0.f;
   if (lane < N) {
-    float ax = 0.f;
-    for (int j = 0; j < M; ++j) ax += L.A[lane * M + j] * L.x[j];
-    ri = L.y[lane] - ax;
+    // 4 independent accumulators: breaks the dependent FMA/LDS chain
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    int j = 0;
+    for (; j + 3 < M; j += 4) {
+      a0 += L.A[lane * M + j + 0] * L.x[j + 0];
+      a1 += L.A[lane * M + j + 1] * L.x[j + 1];
+      a2 += L.A[lane * M + j + 2] * L.x[j + 2];
+      a3 += L.A[lane * M + j + 3] * L.x[j + 3];
+    }
+    for (; j < M; ++j) a0 += L.A[lane * M + j] * L.x[j];
+    ri = L.y[lane] - ((a0 + a1) + (a2 + a3));
     L.r[lane] = ri;
   }
   float xl = (lane < M) ? L.x[lane] : 0.f;
-  float loss = wave_sum(ri * ri) + rho1 * wave_sum(xl * xl)
-               + rho2 * wave_sum(fabsf(xl));
+  float loss = wave_sum(ri * ri + rho1 * xl * xl + rho2 * fabsf(xl));
   __builtin_amdgcn_s_barrier();  // r[] visible (single wave: lockstep, but
                                  // keep an explicit ordering point)
   if (lane < M) {
-    float atr = 0.f;
-    for (int i = 0; i < N; ++i) atr += L.A[i * M + lane] * L.r[i];
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    int i = 0;
+    for (; i + 3 < N; i += 4) {
+      a0 += L.A[(i + 0) * M + lane] * L.r[i + 0];
+      a1 += L.A[(i + 1) * M + lane] * L.r[i + 1];
+      a2 += L.A[(i + 2) * M + lane] * L.r[i + 2];
+      a3 += L.A[(i + 3) * M + lane] * L.r[i + 3];
+    }
+    for (; i < N; ++i) a0 += L.A[i * M + lane] * L.r[i];
+    float atr = (a0 + a1) + (a2 + a3);
     float sgn = (xl > 0.f) ? 1.f : (xl < 0.f ? -1.f : 0.f);
     L.g[lane] = -2.f * atr + 2.f * rho1 * xl + rho2 * sgn;
   }
@@ -255,6 +271,7 @@ extern "C" __global__ __launch_bounds__(64) void enet_lbfgs_solve_kernel(
   L.bg1 = p; p += M;
   L.ro = p; p += HMAX;
   L.al = p; p += HMAX;
+  L.xprev = p; p += M;
 
   for (int i = lane; i < N * M; i += WAVE) L.A[i] = Ag[(long)env * N * M + i];
   if (lane < N) L.y[lane] = yg[(long)env * N + lane];
@@ -270,15 +287,23 @@ extern "C" __global__ __launch_bounds__(64) void enet_lbfgs_solve_kernel(
   int n_iter = 0;
   float t = 1.f, H_diag = 1.f, prev_loss = 0.f;
   float prev_epoch_loss = 1e30f;
+  if (lane < M) L.xprev[lane] = 1e30f;
 
   for (int epoch = 0; epoch < epochs; ++epoch) {
     float loss = eval_loss_grad(L, N, M, rho1, rho2);
     // epoch-level early stop: once the whole optimizer has converged the
     // reference just keeps burning closure evaluations; identical x at
     // fp32, ~3-5x fewer evaluations.
-    if (fabsf(loss - prev_epoch_loss) <= 1e-8f * fmaxf(1.f, fabsf(loss)))
+    if (fabsf(loss - prev_epoch_loss) <= 3e-7f * fmaxf(1.f, fabsf(loss)))
       break;
     prev_epoch_loss = loss;
+    // second stop: parameters no longer moving between epochs
+    {
+      float dx = (lane < M) ? fabsf(L.x[lane] - L.xprev[lane]) : 0.f;
+      float dmax = wave_max(dx);
+      if (lane < M) L.xprev[lane] = L.x[lane];
+      if (dmax <= 1e-7f) break;
+    }
     if (lds_absmax(L.g, M) <= TOL_GRAD) break;
     bool outer_done = false;
     for (int it = 0; it < max_iter; ++it) {
@@ -462,13 +487,18 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
       B[lane * N + r] = v;
     }
 
-  // cyclic Jacobi (values only)
+  // cyclic Jacobi (values only); break on RELATIVE off-diagonal norm
+  float dscale = 0.f;
+  {
+    float dv = (lane < N) ? B[lane * N + lane] : 0.f;
+    dscale = wave_sum(dv * dv) + 1e-30f;
+  }
   for (int sweep = 0; sweep < 12; ++sweep) {
     float off = 0.f;
     for (int pi = 0; pi < N - 1; ++pi)
       if (lane < N && lane > pi) off += B[pi * N + lane] * B[pi * N + lane];
     off = wave_sum(off);
-    if (off < 1e-14f) break;
+    if (off < 1e-12f * dscale) break;
     for (int pi = 0; pi < N - 1; ++pi) {
       for (int q = pi + 1; q < N; ++q) {
         float apq = B[pi * N + q];
