@@ -12,4 +12,5 @@ parsers/writers for the reference's text formats are provided for
 compatibility (`smartcal_amd.radio.sky`, `smartcal_amd.radio.solutions`).
 """
 
-from . import coords, sky, coherency, consensus, hessian, solutions  # noqa: F401
+from . import (coords, sky, coherency, consensus, hessian, solutions,  # noqa: F401
+               array, sim, solver, imaging, influence)  # noqa: F401
