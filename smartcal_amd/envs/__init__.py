@@ -1,1 +1,3 @@
 from .enet import ENetEnv  # noqa: F401
+from .calib import CalibEnv  # noqa: F401
+from .demix import DemixingEnv  # noqa: F401

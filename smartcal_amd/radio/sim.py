@@ -153,7 +153,8 @@ def make_calibration_sky(K: int, rng: np.random.Generator,
             skylmn, rho0.astype(np.float32), ra0, dec0)
 
 
-def make_demixing_sky(rng: np.random.Generator, f0: float = 150e6):
+def make_demixing_sky(rng: np.random.Generator, f0: float = 150e6,
+                      n_outliers: int | None = None):
     """Demixing scenario: 5 A-team outliers + target field (target last,
     as in `generate_data.simulate_data` / `demixingenv.py`). Returns
     (sky, clusters (K=6, target last), separation, azimuth, elevation,
@@ -165,10 +166,11 @@ def make_demixing_sky(rng: np.random.Generator, f0: float = 150e6):
     dec0 = rng.uniform(math.radians(20), math.radians(80))
     lst = ra0 + rng.uniform(-0.3, 0.3)   # target near transit ± a bit
     names, ras, decs, sIs, sPs, clusters = [], [], [], [], [], []
-    K = len(ATEAM) + 1
+    ateam = ATEAM if n_outliers is None else ATEAM[:n_outliers]
+    K = len(ateam) + 1
     sep = np.zeros(K); az = np.zeros(K); el = np.zeros(K)
     fluxes = np.zeros(K)
-    for i, (nm, ra, dec, flux) in enumerate(ATEAM):
+    for i, (nm, ra, dec, flux) in enumerate(ateam):
         names.append(nm); ras.append(ra); decs.append(dec)
         sIs.append(flux); sPs.append(-0.7)
         clusters.append(ClusterDef(i + 1, 1, [nm]))
