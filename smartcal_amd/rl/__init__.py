@@ -1,1 +1,2 @@
 from .buffers import ReplayBuffer, PERBuffer, SumTree  # noqa: F401
+from .buffers_dict import DictReplayBuffer, DictPERBuffer  # noqa: F401

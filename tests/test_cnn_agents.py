@@ -1,0 +1,104 @@
+"""Tests for the CNN (dict-observation) SAC/TD3/DDPG agents."""
+
+import numpy as np
+import pytest
+import torch
+
+from smartcal_amd.rl import sac_cnn, td3_cnn, ddpg_cnn
+
+IMG = (1, 32, 32)
+META = 20
+NACT = 6
+
+
+def _obs(rng):
+    return {"infmap": rng.standard_normal(IMG).astype(np.float32),
+            "metadata": rng.standard_normal(META).astype(np.float32)}
+
+
+def _fill_and_learn(agent, rng, n=12):
+    obs = _obs(rng)
+    for _ in range(n):
+        a = agent.choose_action(obs)
+        assert a.shape == (NACT,)
+        assert np.isfinite(a).all() and (np.abs(a) <= 1.0 + 1e-5).all()
+        obs2 = _obs(rng)
+        agent.store_transition(obs, a, float(rng.standard_normal()), obs2,
+                               False, hint=np.zeros(NACT, np.float32))
+        obs = obs2
+    agent.learn()
+
+
+@pytest.mark.parametrize("prioritized,use_hint", [(False, False),
+                                                  (True, True)])
+def test_sac_cnn_learn(prioritized, use_hint, tmp_path):
+    rng = np.random.default_rng(0)
+    agent = sac_cnn.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=IMG,
+                          batch_size=8, n_actions=NACT, max_mem_size=64,
+                          meta_dim=META, prioritized=prioritized,
+                          use_hint=use_hint, device=torch.device("cpu"),
+                          checkpoint_dir=str(tmp_path))
+    before = agent.actor_fp.flat.clone()
+    _fill_and_learn(agent, rng)
+    assert not torch.equal(before, agent.actor_fp.flat)
+    agent.save_models()
+    agent2 = sac_cnn.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=IMG,
+                           batch_size=8, n_actions=NACT, max_mem_size=64,
+                           meta_dim=META, device=torch.device("cpu"),
+                           checkpoint_dir=str(tmp_path))
+    agent2.load_models()
+    torch.testing.assert_close(agent2.actor_fp.flat, agent.actor_fp.flat)
+
+
+def test_sac_cnn_meta_only():
+    rng = np.random.default_rng(1)
+    agent = sac_cnn.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=IMG,
+                          batch_size=4, n_actions=NACT, max_mem_size=32,
+                          meta_dim=META, use_influence=False,
+                          device=torch.device("cpu"))
+    _fill_and_learn(agent, rng, n=6)
+
+
+def test_td3_cnn_learn(tmp_path):
+    rng = np.random.default_rng(2)
+    agent = td3_cnn.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=IMG,
+                          batch_size=8, n_actions=NACT, max_mem_size=64,
+                          meta_dim=META, warmup=4, use_hint=True,
+                          prioritized=True, device=torch.device("cpu"),
+                          checkpoint_dir=str(tmp_path))
+    before = agent.critic_1_fp.flat.clone()
+    _fill_and_learn(agent, rng)
+    agent.learn()   # second learn hits the delayed actor update
+    assert not torch.equal(before, agent.critic_1_fp.flat)
+    agent.save_models()
+    agent.load_models()
+
+
+def test_ddpg_cnn_learn(tmp_path):
+    rng = np.random.default_rng(3)
+    agent = ddpg_cnn.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3, input_dims=IMG,
+                           batch_size=8, n_actions=NACT, max_mem_size=64,
+                           meta_dim=META, device=torch.device("cpu"),
+                           checkpoint_dir=str(tmp_path))
+    before = agent.actor_fp.flat.clone()
+    _fill_and_learn(agent, rng)
+    assert not torch.equal(before, agent.actor_fp.flat)
+
+
+def test_agent_on_env_obs():
+    """CNN SAC consumes actual CalibEnv observations ('img'/'sky' keys)."""
+    from smartcal_amd.envs.calib import CalibEnv
+    env = CalibEnv(M=3, N_stations=8, Nf=2, Ts=1, Tdelta=4, Ninf=32,
+                   admm_iter=1, poly_order=2, device="cpu", inf_nfreq=1,
+                   seed=4)
+    obs = env.reset()
+    agent = sac_cnn.Agent(gamma=0.99, lr_a=1e-3, lr_c=1e-3,
+                          input_dims=(1, 32, 32), batch_size=4,
+                          n_actions=6, max_mem_size=16, M=3,
+                          device=torch.device("cpu"))
+    a = agent.choose_action(obs)
+    obs2, r, done, info = env.step(a)
+    agent.store_transition(obs, a, r, obs2, done)
+    for _ in range(4):
+        agent.store_transition(obs, a, r, obs2, done)
+    agent.learn()
