@@ -134,8 +134,11 @@ def dsolutions_r(C: torch.Tensor, J: torch.Tensor, N: int,
     AdV = AdV.reshape(8, K, 4 * N, B)
 
     eye = torch.eye(4 * N, dtype=C.dtype, device=dev)
-    A = (Dgrad + _EPS * eye).unsqueeze(1)                  # (K,1,4N,4N)
-    dJ = torch.linalg.solve(A, AdV.permute(1, 0, 2, 3))    # (K,8,4N,B)
+    # inv+GEMM instead of batched solve: ROCm's hipblasCtrsmBatched fails
+    # for complex64 with nrhs ≥ 1024 (B can be 1891 at LOFAR scale), and
+    # the GEMM path keeps MFMA busy; Dgrad is ε-regularized
+    Ainv = torch.linalg.inv(Dgrad + _EPS * eye).unsqueeze(1)  # (K,1,4N,4N)
+    dJ = Ainv @ AdV.permute(1, 0, 2, 3)                       # (K,8,4N,B)
     return dJ.permute(1, 0, 2, 3).contiguous()
 
 

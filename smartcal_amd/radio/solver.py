@@ -220,7 +220,9 @@ def calibrate(vis: VisData, sky, clusters, rho_spectral,
              + alpha * torch.eye(Ne, dtype=torch.complex64, device=dev))
         rhs = part.permute(1, 0, 2, 3, 4, 5).reshape(K, Ne, -1) \
             * rhoZ.view(K, 1, 1)
-        Z = torch.linalg.solve(A, rhs).reshape(K, Ne, Ts, N, 2, 2)
+        # inv+GEMM: complex batched trsm is broken for wide RHS on this
+        # ROCm build (see radio.hessian.dsolutions_r note); Ne is tiny
+        Z = (torch.linalg.inv(A) @ rhs).reshape(K, Ne, Ts, N, 2, 2)
         # ---- J-step with prox toward B_f Z − Y/ρ -----------------------
         BZ = torch.einsum('fe,ektnab->ftknab', Bf_t.to(torch.complex64),
                           Z.permute(1, 0, 2, 3, 4, 5))      # (F,Ts,K,N,2,2)
