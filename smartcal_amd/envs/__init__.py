@@ -1,3 +1,4 @@
 from .enet import ENetEnv  # noqa: F401
 from .calib import CalibEnv  # noqa: F401
 from .demix import DemixingEnv  # noqa: F401
+from .demix_fuzzy import FuzzyDemixingEnv  # noqa: F401
