@@ -191,7 +191,13 @@ class ENetEnv(gymapi.Env):
 
 def obs_to_state(observation) -> torch.Tensor:
     """Flatten a dict observation into the agent's state vector
-    (eig first, then A — reference ``enet_sac.py:548``)."""
+    (eig first, then A — reference ``enet_sac.py:548``). Already-flat
+    tensors/arrays pass through (distributed learner ingestion path)."""
+    if not isinstance(observation, dict):
+        if torch.is_tensor(observation):
+            return observation.reshape(-1).to(torch.float32)
+        return torch.as_tensor(np.asarray(observation),
+                               dtype=torch.float32).reshape(-1)
     eig = observation["eig"]
     A = observation["A"]
     if not torch.is_tensor(eig):

@@ -1,0 +1,164 @@
+"""Multi-process CPU tests (gloo, world_size 2) for the distributed
+layer: learner/actor rounds, DP grad all-reduce, solver freq-sharding."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+N = 6
+M = 6
+OBS_DIM = N + N * M
+NACT = 2
+
+
+def _la_worker(rank, world, port, q):
+    from smartcal_amd.distributed.learner_actor import run_process
+    from smartcal_amd.envs.enet import ENetEnv
+    from smartcal_amd.rl.sac import Agent
+
+    torch.manual_seed(rank)
+    np.random.seed(rank)
+
+    def agent_factory():
+        return Agent(gamma=0.99, batch_size=4, n_actions=NACT, tau=0.005,
+                     max_mem_size=64, input_dims=[OBS_DIM], lr_a=1e-3,
+                     lr_c=1e-3, reward_scale=N, alpha=0.03,
+                     device=torch.device("cpu"))
+
+    def env_factory():
+        return ENetEnv(M, N, device=torch.device("cpu"))
+
+    scores = run_process(rank, world, agent_factory, env_factory,
+                         obs_dim=OBS_DIM, n_actions=NACT, episodes=2,
+                         epochs=1, steps=3, learner_addr="127.0.0.1",
+                         learner_port=port, max_transitions=8,
+                         backend="gloo")
+    if rank == 0:
+        q.put(scores)
+
+
+@pytest.mark.timeout(300)
+def test_learner_actor_round():
+    port = 29531
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_la_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+    assert all(p.exitcode == 0 for p in procs), \
+        [p.exitcode for p in procs]
+    scores = q.get()
+    assert len(scores) == 2          # one mean-reward entry per episode
+    assert all(np.isfinite(s) for s in scores)
+
+
+def _dp_worker(rank, world, port, q):
+    import torch.distributed as dist
+    from smartcal_amd.distributed.dp import allreduce_grad_hook
+    from smartcal_amd.rl.sac import Agent
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(0)        # same init on both ranks
+    np.random.seed(rank)        # different data
+    agent = Agent(gamma=0.99, batch_size=4, n_actions=NACT, tau=0.005,
+                  max_mem_size=32, input_dims=[OBS_DIM], lr_a=1e-3,
+                  lr_c=1e-3, reward_scale=N, alpha=0.03,
+                  device=torch.device("cpu"),
+                  grad_hook=allreduce_grad_hook(world))
+    rng = np.random.default_rng(rank)
+    for _ in range(8):
+        s = rng.standard_normal(OBS_DIM).astype(np.float32)
+        agent.store_transition(torch.from_numpy(s),
+                               rng.standard_normal(NACT).astype(np.float32),
+                               float(rng.standard_normal()),
+                               torch.from_numpy(s), False,
+                               np.zeros(NACT, np.float32))
+    torch.manual_seed(123 + 0)  # identical sampling on both ranks? no —
+    agent.learn()
+    # after one synchronized learn step the actor params must be
+    # IDENTICAL across ranks (same init + averaged grads ⇒ same update)
+    q.put((rank, agent.actor_fp.flat.clone()))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp_allreduce_keeps_replicas_synced():
+    port = 29541
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_dp_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = {}
+    for _ in range(2):
+        r, flat = q.get()
+        out[r] = flat
+    for p in procs:
+        p.join(240)
+    assert all(p.exitcode == 0 for p in procs)
+    # Adam on identical init + identical (averaged) grads ⇒ same params,
+    # provided the batch sampling was also identical. Our agents sample
+    # on-device with the global torch seed; ranks set the same
+    # torch.manual_seed before learn() above.
+    torch.testing.assert_close(out[0], out[1])
+
+
+def _solver_worker(rank, world, port, q):
+    import torch.distributed as dist
+    from smartcal_amd.radio import array as arr, sim, solver
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    rng = np.random.default_rng(3)     # same scenario on both ranks
+    layout = arr.lofar_like_layout(N=6, rng=rng)
+    sky, cs, *_, ra0, dec0 = sim.make_demixing_sky(rng, n_outliers=1)
+    freqs_all = np.linspace(120e6, 160e6, 4)
+    vis = sim.simulate_observation(layout, sky, cs, freqs_all, ra0, dec0,
+                                   Ts=1, Tdelta=4, snr=50.0, rng=rng,
+                                   torch_seed=0)
+    # shard frequencies across the two ranks
+    mine = slice(rank * 2, rank * 2 + 2)
+    shard = sim.VisData(uvw=vis.uvw, freqs=vis.freqs[mine],
+                        data=vis.data[mine], N=vis.N, ra0=ra0, dec0=dec0,
+                        Ts=1, Tdelta=4)
+    sol = solver.calibrate(shard, sky, cs, np.full(2, 5.0, np.float32),
+                           admm_iter=3, poly_order=2, n_sweeps=1,
+                           init_sweeps=4)
+    q.put((rank, sol.Z.cpu(), float(torch.linalg.vector_norm(sol.residual)),
+           float(torch.linalg.vector_norm(shard.data))))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_solver_freq_sharded_consensus():
+    port = 29551
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_solver_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    out = {}
+    for _ in range(2):
+        r, Z, res, dat = q.get()
+        out[r] = (Z, res, dat)
+    for p in procs:
+        p.join(240)
+    assert all(p.exitcode == 0 for p in procs)
+    # consensus: both ranks agree on the global Z
+    torch.testing.assert_close(out[0][0], out[1][0], rtol=1e-4, atol=1e-5)
+    # and each rank's local residual is well below its data power
+    for r in range(2):
+        assert out[r][1] < 0.2 * out[r][2]
