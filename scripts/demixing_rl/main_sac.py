@@ -1,0 +1,63 @@
+"""Demixing SAC (CNN) training.
+
+Reference `demixing_rl/main_sac.py:11-98`: K=6, Ninf=128, metadata
+3K+2=20, K actions, batch 256, mem 16000, lr_a 3e-4; warmup episodes
+with random actions; reward scaling ×10 if > 0; --load resumes nets +
+buffer + scores.
+"""
+
+import argparse
+import pickle
+import sys
+from pathlib import Path
+
+import numpy as np
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[2]))
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
+from smartcal_amd.envs.demix import DemixingEnv
+from smartcal_amd.rl.sac_cnn import Agent
+from smartcal_amd.utils.device import seed_everything
+from _loop import run_training
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seed", default=0, type=int)
+    ap.add_argument("--episodes", default=1000, type=int)
+    ap.add_argument("--steps", default=7, type=int)
+    ap.add_argument("--use_hint", action="store_true", default=False)
+    ap.add_argument("--load", action="store_true", default=False)
+    ap.add_argument("--warmup", default=30, type=int)
+    ap.add_argument("--stations", default=62, type=int)
+    ap.add_argument("--influence", action="store_true", default=False)
+    args = ap.parse_args()
+    seed_everything(args.seed)
+
+    K = 6
+    env = DemixingEnv(K=K, Nf=3, Ninf=128, Tdelta=10,
+                      provide_hint=args.use_hint,
+                      provide_influence=args.influence,
+                      N_stations=args.stations, seed=args.seed)
+    agent = Agent(gamma=0.99, batch_size=256, n_actions=K, tau=0.005,
+                  max_mem_size=16000, input_dims=(1, 128, 128),
+                  meta_dim=3 * K + 2, lr_a=3e-4, lr_c=3e-4,
+                  use_hint=args.use_hint, use_influence=args.influence)
+    if args.load:
+        agent.load_models()
+        try:
+            agent.replaymem.load_checkpoint()
+        except FileNotFoundError:
+            pass
+
+    # reference reward shaping: ×10 if positive (`main_sac.py:70,75`)
+    run_training(env, agent, args.episodes, args.steps,
+                 provide_hint=args.use_hint,
+                 warmup_episodes=args.warmup,
+                 reward_shaping=lambda r: r * 10 if r > 0 else r,
+                 save_every=1)
+
+
+if __name__ == "__main__":
+    main()

@@ -149,6 +149,14 @@ class ENetEnv(gymapi.Env):
         err = torch.norm(self.x0 - self.x).item()
         print(f"{float(self.rho[0]):e} {float(self.rho[1]):e} {err:f}")
 
+    def solution_error(self) -> float:
+        """Relative error of the current solution vs the ground truth
+        (the quantity `enet_eval.py:85-112` compares for RL vs grid
+        search)."""
+        x = self.x.to(self.x0.device)
+        return float(torch.norm(self.x0 - x)
+                     / torch.norm(self.x0).clamp(min=1e-12))
+
     def initsol(self):
         """Solve once with the initial rho (reference ``enetenv.py:197``)."""
         self._observe_y()
