@@ -125,6 +125,48 @@ class Agent:
         if self.grad_hook is not None:
             self.grad_hook(fps)
 
+    def _fork(self, *fns):
+        """Run independent small-GEMM chains concurrently: fns[0] stays
+        on the current stream, each other fn forks onto its own side HIP
+        stream (event fork/join). Stream-aware autograd then also runs
+        their backwards concurrently, and the fork/join pattern is
+        captured into the hipGraph as parallel branches. The twin
+        critics/targets are ~tiny GEMMs at ~3% occupancy each — overlap
+        hides most of their wall time."""
+        if self.device.type != "cuda":
+            return tuple(f() for f in fns)
+        nside = len(fns) - 1
+        if not hasattr(self, "_side_streams"):
+            self._side_streams = []
+            self._fork_evs = []
+            self._join_evs = []
+        while len(self._side_streams) < nside:
+            self._side_streams.append(torch.cuda.Stream())
+            self._fork_evs.append(torch.cuda.Event())
+            self._join_evs.append(torch.cuda.Event())
+        results = [None] * len(fns)
+        for i in range(nside):
+            self._fork_evs[i].record()
+            with torch.cuda.stream(self._side_streams[i]):
+                self._fork_evs[i].wait()
+                results[i + 1] = fns[i + 1]()
+                self._join_evs[i].record()
+        results[0] = fns[0]()
+        for i in range(nside):
+            self._join_evs[i].wait()
+        if not torch.cuda.is_current_stream_capturing():
+            # eager mode: guard side-stream allocations against premature
+            # reuse once consumed on the main stream
+            cs = torch.cuda.current_stream()
+            for r in results[1:]:
+                for t in (r if isinstance(r, (tuple, list)) else (r,)):
+                    if torch.is_tensor(t):
+                        t.record_stream(cs)
+        return tuple(results)
+
+    def _pair(self, f1, f2):
+        return self._fork(f1, f2)
+
     def _learn_body(self, state_batch, new_state_batch, action_batch,
                     reward_batch, terminal_batch, hint_batch, is_w=None):
         """Tensor-only learn step (hipGraph-capturable on GPU): soft target,
@@ -132,8 +174,21 @@ class Agent:
         with torch.no_grad():
             new_actions, new_log_probs = self.actor.sample_normal(
                 new_state_batch, reparameterize=False)
-            q1_t = self.target_critic_1(new_state_batch, new_actions)
-            q2_t = self.target_critic_2(new_state_batch, new_actions)
+
+        def _t1():
+            with torch.no_grad():
+                return self.target_critic_1(new_state_batch, new_actions)
+
+        def _t2():
+            with torch.no_grad():
+                return self.target_critic_2(new_state_batch, new_actions)
+
+        # all four critic-family forwards run concurrently on 4 streams
+        q1, q2, q1_t, q2_t = self._fork(
+            lambda: self.critic_1(state_batch, action_batch),
+            lambda: self.critic_2(state_batch, action_batch),
+            _t1, _t2)
+        with torch.no_grad():
             min_next_target = torch.min(q1_t, q2_t) \
                 - self.alpha * new_log_probs
             # masked_fill (not boolean indexing): same semantics as the
@@ -141,9 +196,6 @@ class Agent:
             min_next_target = min_next_target.masked_fill(terminal_batch,
                                                           0.0)
             new_q_value = reward_batch + self.gamma * min_next_target
-
-        q1 = self.critic_1(state_batch, action_batch)
-        q2 = self.critic_2(state_batch, action_batch)
         if is_w is not None:
             critic_1_loss = (is_w * (q1 - new_q_value).pow(2)).mean()
             critic_2_loss = (is_w * (q2 - new_q_value).pow(2)).mean()
@@ -155,13 +207,21 @@ class Agent:
         self.critic_2_opt.zero_grad()
         critic_loss.backward()
         self._grad_sync([self.critic_1_fp, self.critic_2_fp])
-        self.critic_1_opt.step()
-        self.critic_2_opt.step()
+        self._fork(self.critic_1_opt.step, self.critic_2_opt.step)
 
         actions, log_probs = self.actor.sample_normal(state_batch,
                                                       reparameterize=True)
-        q1_pi = self.critic_1(state_batch, actions)
-        q2_pi = self.critic_2(state_batch, actions)
+        # actor update: critic-weight grads from this loss are discarded
+        # (zeroed before the next critic regression), so freeze critic
+        # params here — backward then only propagates through `actions`,
+        # skipping all dW GEMMs of both critics
+        for p in self.critic_1.parameters():
+            p.requires_grad_(False)
+        for p in self.critic_2.parameters():
+            p.requires_grad_(False)
+        q1_pi, q2_pi = self._pair(
+            lambda: self.critic_1(state_batch, actions),
+            lambda: self.critic_2(state_batch, actions))
         critic_value = torch.min(q1_pi, q2_pi)
 
         actor_loss = (self.alpha * log_probs - critic_value).mean()
@@ -174,6 +234,10 @@ class Agent:
                 + self.rho * gfun
         self.actor_opt.zero_grad()
         actor_loss.backward()
+        for p in self.critic_1.parameters():
+            p.requires_grad_(True)
+        for p in self.critic_2.parameters():
+            p.requires_grad_(True)
         self._grad_sync([self.actor_fp])
         self.actor_opt.step()
         self.update_network_parameters()
