@@ -181,3 +181,23 @@ def test_simulate_systematic_errors():
     a = gs[:, :, 0].T       # (8N·Ts, K)
     J = solutions.solutions_to_J(a, N, Ts)
     assert J.shape == (K, 2 * N * Ts, 2)
+
+
+def test_shapelet_roundtrip():
+    from smartcal_amd.radio import shapelet
+    rng = np.random.default_rng(0)
+    text, n0, beta, coeff, ptext = shapelet.generate_random_shapelet_model(
+        rng, perturbed=True)
+    pos, n0b, betab, coeffb = shapelet.parse_shapelet_model(text)
+    assert n0b == n0 and abs(betab - beta) < 1e-12
+    np.testing.assert_allclose(coeffb, coeff, rtol=1e-10)
+    _, n0p, betap, coeffp = shapelet.parse_shapelet_model(ptext)
+    assert betap >= beta
+    # perturbation is ~10% in norm
+    dn = np.linalg.norm(coeffp - coeff) / np.linalg.norm(coeff)
+    assert 0.05 < dn < 0.15
+    # basis: zeroth mode is a normalized Gaussian, orthonormal-ish
+    l = np.linspace(-0.5, 0.5, 101)
+    basis = shapelet.shapelet_basis(3, 0.1, l, np.zeros_like(l))
+    assert basis.shape == (9, 101)
+    assert np.argmax(basis[0]) == 50
