@@ -60,9 +60,10 @@ def train(algo: str, seed: int):
 
 
 def main():
+    seeds = (1, 2, 3)
     out = {}
     for algo in ("sac", "td3", "ddpg"):
-        out[algo] = train(algo, seed=SEED)
+        out[algo] = {str(s): train(algo, seed=s) for s in seeds}
     Path("gpurun_out").mkdir(exist_ok=True)
     with open("gpurun_out/learning_curves.json", "w") as f:
         json.dump(out, f)
@@ -70,13 +71,18 @@ def main():
     matplotlib.use("Agg")
     import matplotlib.pyplot as plt
     plt.figure(figsize=(7, 4.5))
-    for algo, scores in out.items():
-        s = np.asarray(scores)
-        smooth = np.convolve(s, np.ones(25) / 25, mode="valid")
-        plt.plot(smooth, label=algo.upper())
+    colors = {"sac": "C0", "td3": "C1", "ddpg": "C2"}
+    for algo, runs in out.items():
+        sm = np.stack([np.convolve(np.asarray(s), np.ones(25) / 25,
+                                   mode="valid") for s in runs.values()])
+        med = np.median(sm, axis=0)
+        plt.plot(med, color=colors[algo], label=algo.upper())
+        plt.fill_between(np.arange(sm.shape[1]), sm.min(0), sm.max(0),
+                         color=colors[algo], alpha=0.15)
     plt.xlabel("episode")
-    plt.ylabel(f"score (25-episode moving average)")
-    plt.title(f"Elastic-net tuning on 1x MI355X (N=M={N}, batch 64)")
+    plt.ylabel("score (25-episode moving average)")
+    plt.title(f"Elastic-net tuning on 1x MI355X "
+              f"(N=M={N}, batch 64; median of {len(seeds)} seeds)")
     plt.legend()
     plt.grid(alpha=0.3)
     plt.tight_layout()
