@@ -87,10 +87,13 @@ def hessianres(R: torch.Tensor, C: torch.Tensor, J: torch.Tensor,
     Jp = Jv[:, p_idx]                                     # (K,B,2,2)
     Jq = Jv[:, q_idx]
     CiT = Ci.reshape(K, T, B, 2, 2)
-    R1 = CiT @ Jq.mH.unsqueeze(1)
-    D1 = (R1 @ R1.mH).sum(dim=1)                          # (K,B,2,2)
-    R2 = Jp.unsqueeze(1) @ CiT
-    D2 = (R2.mH @ R2).sum(dim=1)
+    # elementwise 2×2 products (radio.small_complex): batched tiny
+    # complex GEMMs are pathological for rocBLAS
+    from .small_complex import mm2, mm2H, Hmm2
+    R1 = mm2H(CiT, Jq.unsqueeze(1))
+    D1 = mm2H(R1, R1).sum(dim=1)                          # (K,B,2,2)
+    R2 = mm2(Jp.unsqueeze(1), CiT)
+    D2 = Hmm2(R2, R2).sum(dim=1)
 
     Hb = C.new_zeros(K, N * N, 4, 4)
     Hb.index_add_(1, p_idx * N + q_idx, Imp)
@@ -120,7 +123,8 @@ def dsolutions_r(C: torch.Tensor, J: torch.Tensor, N: int,
     Ci = _c22(C)
     Jv = J.reshape(K, N, 2, 2)
     Jq = Jv[:, q_idx]                                      # (K,B,2,2)
-    lhs = Jq.unsqueeze(1) @ Ci.reshape(K, T, B, 2, 2).mH   # (K,T,B,2,2)
+    from .small_complex import mm2H
+    lhs = mm2H(Jq.unsqueeze(1), Ci.reshape(K, T, B, 2, 2))  # (K,T,B,2,2)
     lhsT = lhs.mT.sum(dim=1)                               # (K,B,2,2)
 
     Vr = _r_vectors(dev, C.dtype)                          # (8,2,2)
@@ -152,11 +156,32 @@ def _dres_blocks(C, J, N, dJ):
     Ci = _c22(C)
     Jv = J.reshape(K, N, 2, 2)
     Jq = Jv[:, q_idx]
-    lhs = -(Ci.reshape(K, T, B, 2, 2) @ Jq.mH.unsqueeze(1)).mT
+    from .small_complex import mm2H
+    lhs = -(mm2H(Ci.reshape(K, T, B, 2, 2), Jq.unsqueeze(1))).mT
     lhs_sum = lhs.sum(dim=1)                               # (K,B,2,2)
     dJv = dJ.reshape(8, K, 2, N, 2, B)
     gath = dJv[:, :, :, p_idx]                             # (8,K,2,B,2,B)
     return lhs_sum, gath, B, T, K
+
+
+def _dres_contract(lhs_sum, gath, per_k: bool):
+    """Σ_m (and optionally Σ_k) lhs_sum[k,b,i,m]·gath[r,k,m,b,j,c] as
+    2·K broadcasted fused multiplies — the einsum form lowers to M=2
+    batched GEMMs that run ~800 µs each on rocBLAS (the single biggest
+    cost of the whole influence pipeline before this rewrite)."""
+    K, B = lhs_sum.shape[0], lhs_sum.shape[1]
+    Bc = gath.shape[-1]
+    shape = (8, K, B, 2, 2, Bc) if per_k else (8, B, 2, 2, Bc)
+    out = gath.new_zeros(shape)
+    for k in range(K):
+        for m in range(2):
+            term = lhs_sum[k, :, :, m].reshape(1, B, 2, 1, 1) \
+                * gath[:, k, m].reshape(8, B, 1, 2, Bc)
+            if per_k:
+                out[:, k] += term
+            else:
+                out += term
+    return out
 
 
 def dresiduals_rk(C: torch.Tensor, J: torch.Tensor, N: int,
@@ -164,7 +189,7 @@ def dresiduals_rk(C: torch.Tensor, J: torch.Tensor, N: int,
     """dR (8, K, 4B, B): residual derivatives per direction. Batched
     `Dresiduals_rk` (`calibration_tools.py:1129-1178`)."""
     lhs_sum, gath, B, T, K = _dres_blocks(C, J, N, dJ)
-    blocks = torch.einsum('kbim,rkmbjc->rkbijc', lhs_sum, gath)
+    blocks = _dres_contract(lhs_sum, gath, per_k=True)
     if addself:
         Vr = _r_vectors(C.device, C.dtype)                 # (8,2,2)
         ar = torch.arange(B, device=C.device)
@@ -179,7 +204,7 @@ def dresiduals_r(C: torch.Tensor, J: torch.Tensor, N: int,
     """dR (8, 4B, B), summed over directions. Batched
     `Dresiduals_r_torch` (`calibration_tools.py:1078-1126`)."""
     lhs_sum, gath, B, T, K = _dres_blocks(C, J, N, dJ)
-    blocks = torch.einsum('kbim,rkmbjc->rbijc', lhs_sum, gath)
+    blocks = _dres_contract(lhs_sum, gath, per_k=False)
     if addself:
         Vr = _r_vectors(C.device, C.dtype)
         ar = torch.arange(B, device=C.device)
@@ -205,7 +230,8 @@ def log_likelihood_ratio(R: torch.Tensor, C: torch.Tensor, J: torch.Tensor,
     Jv = J.reshape(K, N, 2, 2)
     Jp = Jv[:, p_idx].unsqueeze(1).expand(K, T, -1, 2, 2).reshape(K, S, 2, 2)
     Jq = Jv[:, q_idx].unsqueeze(1).expand(K, T, -1, 2, 2).reshape(K, S, 2, 2)
-    Mu = Jp @ Ci @ Jq.mH                                   # (K,S,2,2)
+    from .small_complex import mm2, mm2H
+    Mu = mm2H(mm2(Jp, Ci), Jq)                             # (K,S,2,2)
     rn = (Res.abs() ** 2).sum()
     rmn = ((Res.unsqueeze(0) + Mu).abs() ** 2).sum(dim=(1, 2, 3))
     return ((rmn - rn) / (sigma2 + _EPS)).to(torch.float32)

@@ -217,7 +217,8 @@ def apply_jones(C: torch.Tensor, J: torch.Tensor, N: int,
     C22 = C.reshape(K, T, B, 2, 2)
     Jp = J[:, t_int][:, :, p_idx]                     # (K,T,B,2,2)
     Jq = J[:, t_int][:, :, q_idx]
-    V = (Jp @ C22 @ Jq.mH).sum(dim=0)                 # (T,B,2,2)
+    from .small_complex import mm2, mm2H
+    V = mm2H(mm2(Jp, C22), Jq).sum(dim=0)             # (T,B,2,2)
     return V.reshape(S, 4)
 
 

@@ -96,17 +96,20 @@ def _solve_sweeps(data22, C22, J, rho_t, prox_target, p_idx, q_idx, N,
         Jq = Jt[:, :, :, q_idx]                            # (F,T,K,B,2,2)
         Jp = Jt[:, :, :, p_idx]
         Cp = C22.permute(0, 2, 3, 1, 4, 5)                 # (F,T,B,K,2,2)
-        A_p = Cp @ Jq.permute(0, 1, 3, 2, 4, 5).mH         # (F,T,B,K,2,2)
-        A_q = Cp.mH @ Jp.permute(0, 1, 3, 2, 4, 5).mH
+        # elementwise small-complex products (see radio.small_complex):
+        # batched 2×2 / k=2 shapes are pathological for rocBLAS
+        from .small_complex import mm2H, Hmm2, abH_k2
+        A_p = mm2H(Cp, Jq.permute(0, 1, 3, 2, 4, 5))       # (F,T,B,K,2,2)
+        A_q = Hmm2(Cp, Jp.permute(0, 1, 3, 2, 4, 5).conj().mT)
         V = data22                                          # (F,T,B,2,2)
         # accumulate per (station, interval): normal matrix (2K,2K), rhs (2,2K)
         # W = A stacked over k → (2K,2) ; contribution V·W^H (2,2K), W·W^H
         Wp = A_p.reshape(F, T, Bn, 2 * K, 2)
         Wq = A_q.reshape(F, T, Bn, 2 * K, 2)
-        rhs_p = V @ Wp.mH                                   # (F,T,B,2,2K)
-        rhs_q = V.mH @ Wq.mH
-        nm_p = Wp @ Wp.mH                                   # (F,T,B,2K,2K)
-        nm_q = Wq @ Wq.mH
+        rhs_p = abH_k2(V, Wp)                               # (F,T,B,2,2K)
+        rhs_q = abH_k2(V.mH, Wq)
+        nm_p = abH_k2(Wp, Wp)                               # (F,T,B,2K,2K)
+        nm_q = abH_k2(Wq, Wq)
         # scatter-add into (F,Ts,N,…) by interval and station
         rhs = V.new_zeros(F, Ts, N, 2, 2 * K)
         nm = V.new_zeros(F, Ts, N, 2 * K, 2 * K)
