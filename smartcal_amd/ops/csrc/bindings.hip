@@ -22,8 +22,9 @@ extern "C" __global__ void ln_act_bwd_kernel(const float*, const float*,
 extern "C" __global__ void gemm_f32_nn_kernel(const float*, const float*,
                                               float*, int, int, int);
 extern "C" __global__ void gemm_f32_tn_kernel(const float*, const float*,
-                                              float*, int, int, int);
-extern "C" __global__ void colsum_kernel(const float*, float*, int, int);
+                                              float*, int, int, int, int);
+extern "C" __global__ void colsum_kernel(const float*, float*, int, int,
+                                         int);
 extern "C" __global__ void tanh_gauss_fwd_kernel(const float*, const float*,
                                                  const float*, float*, float*,
                                                  float*, float, int, int);
@@ -103,6 +104,28 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_bwd_dz(
   return {dz, dgamma, dbeta};
 }
 
+// Direct-accumulate variant of fused_linear_bwd_dz: dgamma/dbeta are
+// (pre-zeroed) flat-grad views accumulated atomically by the kernel.
+at::Tensor fused_linear_bwd_dz_into(
+    const at::Tensor& dy, const at::Tensor& y, const at::Tensor& zhat,
+    const at::Tensor& rstd, const c10::optional<at::Tensor>& gamma,
+    int64_t act, bool with_ln, at::Tensor dgamma, at::Tensor dbeta) {
+  check_f32(dy, "dy");
+  const int B = dy.size(0), N = dy.size(1);
+  auto dz = at::empty({B, N}, dy.options());
+  TORCH_CHECK(dgamma.is_contiguous() && dbeta.is_contiguous(),
+              "grad views must be contiguous");
+  hipLaunchKernelGGL(ln_act_bwd_kernel, dim3(B), dim3(256), 0, stream(),
+                     dy.data_ptr<float>(), y.data_ptr<float>(),
+                     with_ln ? zhat.data_ptr<float>() : nullptr,
+                     with_ln ? rstd.data_ptr<float>() : nullptr,
+                     gamma ? gamma->data_ptr<float>() : nullptr,
+                     dz.data_ptr<float>(), dgamma.data_ptr<float>(),
+                     dbeta.data_ptr<float>(), B, N, (int)act,
+                     with_ln ? 1 : 0);
+  return dz;
+}
+
 at::Tensor mfma_gemm_nn(const at::Tensor& A, const at::Tensor& B) {
   check_f32(A, "A");
   check_f32(B, "B");
@@ -127,11 +150,31 @@ std::tuple<at::Tensor, at::Tensor> mfma_gemm_tn_bias(const at::Tensor& dz,
   dim3 grid((N + 15) / 16, (K + 63) / 64);
   hipLaunchKernelGGL(gemm_f32_tn_kernel, grid, dim3(256), 0, stream(),
                      dz.data_ptr<float>(), x.data_ptr<float>(),
-                     dW.data_ptr<float>(), N, Bb, K);
+                     dW.data_ptr<float>(), N, Bb, K, 0);
   hipLaunchKernelGGL(colsum_kernel, dim3((N + 63) / 64), dim3(256), 0,
                      stream(), dz.data_ptr<float>(), db.data_ptr<float>(),
-                     Bb, N);
+                     Bb, N, 0);
   return {dW, db};
+}
+
+// Direct-accumulate variant: dW/db are views into the (pre-zeroed) flat
+// gradient pool — the GEMM and colsum write with += and the caller skips
+// autograd's per-parameter zero+add kernels entirely.
+void mfma_gemm_tn_bias_into(const at::Tensor& dz, const at::Tensor& x,
+                            at::Tensor dW, at::Tensor db) {
+  check_f32(dz, "dz");
+  check_f32(x, "x");
+  const int Bb = dz.size(0), N = dz.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == Bb, "gemm_tn batch mismatch");
+  TORCH_CHECK(dW.is_contiguous() && db.is_contiguous(),
+              "grad views must be contiguous");
+  dim3 grid((N + 15) / 16, (K + 63) / 64);
+  hipLaunchKernelGGL(gemm_f32_tn_kernel, grid, dim3(256), 0, stream(),
+                     dz.data_ptr<float>(), x.data_ptr<float>(),
+                     dW.data_ptr<float>(), N, Bb, K, 1);
+  hipLaunchKernelGGL(colsum_kernel, dim3((N + 63) / 64), dim3(256), 0,
+                     stream(), dz.data_ptr<float>(), db.data_ptr<float>(),
+                     Bb, N, 1);
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> tanh_gauss_fwd(
@@ -235,6 +278,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_linear_bwd_dz", &fused_linear_bwd_dz);
   m.def("mfma_gemm_nn", &mfma_gemm_nn);
   m.def("mfma_gemm_tn_bias", &mfma_gemm_tn_bias);
+  m.def("mfma_gemm_tn_bias_into", &mfma_gemm_tn_bias_into);
+  m.def("fused_linear_bwd_dz_into", &fused_linear_bwd_dz_into);
   m.def("tanh_gauss_fwd", &tanh_gauss_fwd);
   m.def("tanh_gauss_bwd", &tanh_gauss_bwd);
   m.def("fused_adam", &fused_adam);

@@ -56,9 +56,11 @@ extern "C" __global__ __launch_bounds__(256) void gemm_f32_nn_kernel(
 }
 
 // C (M,N) = A^T @ B with A (K,M), B (K,N) row-major (e.g. dW = dZ^T @ X).
+// accumulate: C += result (direct-accumulate into a pre-zeroed flat grad
+// pool — removes the separate autograd add/zero kernels per parameter).
 extern "C" __global__ __launch_bounds__(256) void gemm_f32_tn_kernel(
     const float* __restrict__ A, const float* __restrict__ B,
-    float* __restrict__ C, int M, int K, int N) {
+    float* __restrict__ C, int M, int K, int N, int accumulate) {
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -81,15 +83,19 @@ extern "C" __global__ __launch_bounds__(256) void gemm_f32_tn_kernel(
   for (int r = 0; r < 4; ++r) {
     const int orow = row0 + l4 * 4 + r;
     const int ocol = col0 + l15;
-    if (orow < M && ocol < N) C[(long)orow * N + ocol] = acc[r];
+    if (orow < M && ocol < N) {
+      const long idx = (long)orow * N + ocol;
+      C[idx] = accumulate ? C[idx] + acc[r] : acc[r];
+    }
   }
 }
 
 // db (N) = column sums of DZ (B, N). Block = 256 threads covering 64
 // columns x 4 row-stripes (coalesced: consecutive threads -> consecutive
-// columns), combined through LDS.
+// columns), combined through LDS. accumulate: OUT += sums.
 extern "C" __global__ __launch_bounds__(256) void colsum_kernel(
-    const float* __restrict__ DZ, float* __restrict__ OUT, int B, int N) {
+    const float* __restrict__ DZ, float* __restrict__ OUT, int B, int N,
+    int accumulate) {
   __shared__ float red[4 * 64];
   const int c = blockIdx.x * 64 + (threadIdx.x & 63);
   const int stripe = threadIdx.x >> 6;
@@ -98,7 +104,9 @@ extern "C" __global__ __launch_bounds__(256) void colsum_kernel(
     for (int r = stripe; r < B; r += 4) s += DZ[(long)r * N + c];
   red[stripe * 64 + (threadIdx.x & 63)] = s;
   __syncthreads();
-  if (threadIdx.x < 64 && c < N)
-    OUT[c] = red[threadIdx.x] + red[64 + threadIdx.x] +
-             red[128 + threadIdx.x] + red[192 + threadIdx.x];
+  if (threadIdx.x < 64 && c < N) {
+    const float s4 = red[threadIdx.x] + red[64 + threadIdx.x] +
+                     red[128 + threadIdx.x] + red[192 + threadIdx.x];
+    OUT[c] = accumulate ? OUT[c] + s4 : s4;
+  }
 }

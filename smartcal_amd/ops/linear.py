@@ -34,12 +34,29 @@ _ACT_CODES = {"none": ACT_NONE, "elu": ACT_ELU, "relu": ACT_RELU,
               "tanh": ACT_TANH}
 
 
+def _grad_view(p: torch.Tensor) -> torch.Tensor:
+    """The tensor the kernels accumulate this parameter's gradient into.
+
+    Agents bind each parameter's .grad to a slice of one flat pool
+    (``utils.flatten.FlatParams``), so the backward kernels can write
+    with += directly — no per-parameter autograd zero/add kernels in the
+    learn graph. Standalone tensors (tests) get a zeroed .grad created
+    on first use, which matches autograd's accumulate semantics.
+    """
+    if p.grad is None:
+        p.grad = torch.zeros_like(p)
+    return p.grad
+
+
 class _FusedLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, W, b, gamma, beta, act_code: int, with_ln: bool):
         y, zhat, rstd = ext().fused_linear_fwd(x, W, b, gamma, beta,
                                                act_code, with_ln)
         ctx.save_for_backward(x, W, gamma, zhat, rstd, y)
+        # python refs to the parameter objects for direct grad
+        # accumulation (not part of the autograd graph)
+        ctx.param_refs = (W, b, gamma, beta)
         ctx.act_code = act_code
         ctx.with_ln = with_ln
         return y
@@ -47,14 +64,26 @@ class _FusedLinearFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, W, gamma, zhat, rstd, y = ctx.saved_tensors
+        Wp, bp, gp, bep = ctx.param_refs
         dy = dy.contiguous()
-        dz, dgamma, dbeta = ext().fused_linear_bwd_dz(
-            dy, y, zhat, rstd, gamma, ctx.act_code, ctx.with_ln)
+        want_w = ctx.needs_input_grad[1]
+        if ctx.with_ln and want_w:
+            dz = ext().fused_linear_bwd_dz_into(
+                dy, y, zhat, rstd, gamma, ctx.act_code, True,
+                _grad_view(gp), _grad_view(bep))
+        else:
+            dz, dgamma, dbeta = ext().fused_linear_bwd_dz(
+                dy, y, zhat, rstd, gamma, ctx.act_code, ctx.with_ln)
+            if ctx.with_ln and want_w:
+                _grad_view(gp).add_(dgamma)
+                _grad_view(bep).add_(dbeta)
         dx = ext().mfma_gemm_nn(dz, W)          # (B,N) @ (N,K) -> (B,K)
-        dW, db = ext().mfma_gemm_tn_bias(dz, x)  # dz^T @ x -> (N,K), col-sum dz
-        dgamma_out = dgamma if ctx.with_ln else None
-        dbeta_out = dbeta if ctx.with_ln else None
-        return dx, dW, db, dgamma_out, dbeta_out, None, None
+        if want_w:
+            # dW = dz^T @ x and db = col-sum(dz), accumulated straight
+            # into the flat gradient pool by the kernels
+            ext().mfma_gemm_tn_bias_into(dz, x, _grad_view(Wp),
+                                         _grad_view(bp))
+        return dx, None, None, None, None, None, None
 
 
 def fused_linear(x: torch.Tensor, W: torch.Tensor,
