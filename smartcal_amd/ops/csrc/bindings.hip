@@ -104,6 +104,83 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_bwd_dz(
   return {dz, dgamma, dbeta};
 }
 
+struct ChainArgs {
+  const float* W[4];
+  const float* bias[4];
+  const float* gamma[4];
+  const float* beta[4];
+  float* Y[4];
+  float* ZHAT[4];
+  float* RSTD[4];
+  int dims[5];
+  int act[4];
+  int with_ln[4];
+  int L;
+  int B;
+};
+extern "C" __global__ void mlp_chain_fwd_kernel(const float*, ChainArgs);
+
+// One kernel for a whole Linear(+LN)(+act) chain (≤4 layers, widths ≤512):
+// activations ping-pong in LDS; per-layer y/zhat/rstd still stored for
+// backward. Returns lists [y_l], [zhat_l], [rstd_l].
+std::tuple<std::vector<at::Tensor>, std::vector<at::Tensor>,
+           std::vector<at::Tensor>>
+mlp_chain_fwd(const at::Tensor& x,
+              const std::vector<at::Tensor>& Ws,
+              const std::vector<at::Tensor>& bs,
+              const std::vector<at::Tensor>& gammas,
+              const std::vector<at::Tensor>& betas,
+              const std::vector<int64_t>& acts) {
+  check_f32(x, "x");
+  const int L = (int)Ws.size();
+  TORCH_CHECK(L >= 1 && L <= 4, "chain supports 1..4 layers");
+  const int B = x.size(0);
+  ChainArgs args{};
+  args.L = L;
+  args.B = B;
+  args.dims[0] = x.size(1);
+  std::vector<at::Tensor> ys, zhats, rstds;
+  for (int l = 0; l < L; ++l) {
+    check_f32(Ws[l], "W");
+    const int N = Ws[l].size(0);
+    TORCH_CHECK(Ws[l].size(1) == args.dims[l], "chain dim mismatch");
+    TORCH_CHECK(N <= 512 && args.dims[l] <= 512, "chain widths <= 512");
+    args.dims[l + 1] = N;
+    args.W[l] = Ws[l].data_ptr<float>();
+    args.bias[l] = bs[l].defined() ? bs[l].data_ptr<float>() : nullptr;
+    const bool ln = gammas[l].defined();
+    args.with_ln[l] = ln ? 1 : 0;
+    args.gamma[l] = ln ? gammas[l].data_ptr<float>() : nullptr;
+    args.beta[l] = ln ? betas[l].data_ptr<float>() : nullptr;
+    args.act[l] = (int)acts[l];
+    ys.push_back(at::empty({B, N}, x.options()));
+    zhats.push_back(ln ? at::empty({B, N}, x.options())
+                       : at::empty({0}, x.options()));
+    rstds.push_back(ln ? at::empty({B}, x.options())
+                       : at::empty({0}, x.options()));
+    args.Y[l] = ys[l].data_ptr<float>();
+    args.ZHAT[l] = ln ? zhats[l].data_ptr<float>() : nullptr;
+    args.RSTD[l] = ln ? rstds[l].data_ptr<float>() : nullptr;
+  }
+  // dynamic LDS: 2 activation buffers + per-wave W subtiles (16 waves)
+  // + row stats — ~139 KB of the 160 KB LDS (one block per CU; only
+  // ceil(B/16) blocks exist anyway)
+  const int XP = 577;
+  const int lds_bytes =
+      (2 * 16 * XP + 16 * 16 * 65 + 16 * 16 * 2 + 16 * 2)
+      * (int)sizeof(float);
+  static bool attr_set = false;
+  if (!attr_set) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&mlp_chain_fwd_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+    attr_set = true;
+  }
+  hipLaunchKernelGGL(mlp_chain_fwd_kernel, dim3((B + 15) / 16), dim3(1024),
+                     lds_bytes, stream(), x.data_ptr<float>(), args);
+  return {ys, zhats, rstds};
+}
+
 // Direct-accumulate variant of fused_linear_bwd_dz: dgamma/dbeta are
 // (pre-zeroed) flat-grad views accumulated atomically by the kernel.
 at::Tensor fused_linear_bwd_dz_into(
@@ -280,6 +357,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_gemm_tn_bias", &mfma_gemm_tn_bias);
   m.def("mfma_gemm_tn_bias_into", &mfma_gemm_tn_bias_into);
   m.def("fused_linear_bwd_dz_into", &fused_linear_bwd_dz_into);
+  m.def("mlp_chain_fwd", &mlp_chain_fwd);
   m.def("tanh_gauss_fwd", &tanh_gauss_fwd);
   m.def("tanh_gauss_bwd", &tanh_gauss_bwd);
   m.def("fused_adam", &fused_adam);

@@ -151,3 +151,71 @@ class FusedLinear(torch.nn.Module):
     def extra_repr(self):
         return (f"in={self.in_features}, out={self.out_features}, "
                 f"ln={self.ln_weight is not None}, act={self.act}")
+
+
+class _FusedChainFn(torch.autograd.Function):
+    """Whole-chain forward (one kernel) with per-layer direct-accumulate
+    backward. flat params layout: [W, b, gamma, beta] per layer."""
+
+    @staticmethod
+    def forward(ctx, x, acts, *params):
+        L = len(acts)
+        Ws = [params[4 * l + 0] for l in range(L)]
+        bs = [params[4 * l + 1] for l in range(L)]
+        gs = [params[4 * l + 2] for l in range(L)]
+        bes = [params[4 * l + 3] for l in range(L)]
+        ys, zhats, rstds = ext().mlp_chain_fwd(
+            x, Ws, bs, gs, bes, [int(a) for a in acts])
+        ctx.save_for_backward(x, *ys, *zhats, *rstds)
+        ctx.param_refs = params
+        ctx.acts = acts
+        ctx.L = L
+        return ys[-1]
+
+    @staticmethod
+    def backward(ctx, dy):
+        L = ctx.L
+        saved = ctx.saved_tensors
+        x = saved[0]
+        ys = saved[1:1 + L]
+        zhats = saved[1 + L:1 + 2 * L]
+        rstds = saved[1 + 2 * L:1 + 3 * L]
+        params = ctx.param_refs
+        dy = dy.contiguous()
+        for l in range(L - 1, -1, -1):
+            W, b, g, be = params[4 * l:4 * l + 4]
+            want_w = ctx.needs_input_grad[2 + 4 * l]
+            if want_w:
+                dz = ext().fused_linear_bwd_dz_into(
+                    dy, ys[l], zhats[l], rstds[l], g, ctx.acts[l], True,
+                    _grad_view(g), _grad_view(be))
+            else:
+                dz, _, _ = ext().fused_linear_bwd_dz(
+                    dy, ys[l], zhats[l], rstds[l], g, ctx.acts[l], True)
+            x_l = x if l == 0 else ys[l - 1]
+            if want_w:
+                ext().mfma_gemm_tn_bias_into(dz, x_l, _grad_view(W),
+                                             _grad_view(b))
+            dy = ext().mfma_gemm_nn(dz, W)
+        return (dy, None) + (None,) * (4 * L)
+
+
+def fused_chain(x: torch.Tensor, layers) -> torch.Tensor:
+    """Run a stack of LN+act FusedLinear layers as ONE kernel on GPU
+    (activations stay in LDS between layers); CPU falls back to the
+    per-layer path."""
+    squeeze = x.dim() == 1
+    if squeeze:
+        x = x.unsqueeze(0)
+    if use_hip(x) and len(layers) >= 2 \
+            and all(m.ln_weight is not None for m in layers):
+        acts = tuple(_ACT_CODES[m.act] for m in layers)
+        params = []
+        for m in layers:
+            params += [m.weight, m.bias, m.ln_weight, m.ln_bias]
+        y = _FusedChainFn.apply(x.contiguous(), acts, *params)
+    else:
+        y = x
+        for m in layers:
+            y = m(y)
+    return y.squeeze(0) if squeeze else y

@@ -19,7 +19,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops.linear import FusedLinear, fused_linear
+from ..ops.linear import FusedLinear, fused_linear, fused_chain
 from ..ops.sampling import tanh_gauss_sample
 
 LOGSIG_MIN = -20.0
@@ -38,7 +38,8 @@ class SACActorMLP(nn.Module):
                                 init_scale=0.003)
 
     def forward(self, x: torch.Tensor):
-        h = self.l3(self.l2(self.l1(x)))
+        # whole hidden stack as ONE chain kernel (activations in LDS)
+        h = fused_chain(x, (self.l1, self.l2, self.l3))
         out = self.head(h)
         mu = out[..., :self.n_actions]
         logsigma = out[..., self.n_actions:].clamp(LOGSIG_MIN, LOGSIG_MAX)
@@ -65,7 +66,7 @@ class DeterministicActorMLP(nn.Module):
                                 init_scale=0.003)
 
     def forward(self, x: torch.Tensor):
-        return self.head(self.l3(self.l2(self.l1(x))))
+        return self.head(fused_chain(x, (self.l1, self.l2, self.l3)))
 
 
 class CriticMLP(nn.Module):
@@ -79,7 +80,7 @@ class CriticMLP(nn.Module):
                                 init_scale=0.003)
 
     def forward(self, state: torch.Tensor, action: torch.Tensor):
-        x = self.s2(self.s1(state))
-        y = self.a2(self.a1(action))
+        x = fused_chain(state, (self.s1, self.s2))
+        y = fused_chain(action, (self.a1, self.a2))
         z = torch.cat((x, y), dim=-1)
         return self.head(z)

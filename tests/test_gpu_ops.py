@@ -262,3 +262,39 @@ def test_sac_agent_gpu_learn():
     torch.cuda.synchronize()
     assert agent.learn_counter >= 1
     assert torch.isfinite(agent.actor_fp.flat).all()
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason='needs GPU')
+def test_mlp_chain_matches_layers():
+    """Chain kernel ≡ per-layer fused path: outputs and all grads."""
+    import copy
+    from smartcal_amd.ops.linear import FusedLinear, fused_chain
+    torch.manual_seed(0)
+    layers = torch.nn.ModuleList([
+        FusedLinear(420, 512), FusedLinear(512, 256),
+        FusedLinear(256, 128)]).cuda()
+    layers_ref = copy.deepcopy(layers)
+    for B in (1, 64):
+        x = torch.randn(B, 420, device="cuda")
+        y = fused_chain(x, tuple(layers))
+        y_ref = x
+        for m in layers_ref:
+            y_ref = m(y_ref)
+        torch.testing.assert_close(y, y_ref, rtol=2e-4, atol=2e-4)
+    # gradients
+    for m in (*layers, *layers_ref):
+        for p in m.parameters():
+            p.grad = None
+    x = torch.randn(64, 420, device="cuda", requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    fused_chain(x, tuple(layers)).square().sum().backward()
+    yr = x2
+    for m in layers_ref:
+        yr = m(yr)
+    yr.square().sum().backward()
+    torch.testing.assert_close(x.grad, x2.grad, rtol=2e-3, atol=2e-3)
+    for m, mr in zip(layers, layers_ref):
+        for (n, p), (_, pr) in zip(m.named_parameters(),
+                                   mr.named_parameters()):
+            torch.testing.assert_close(p.grad, pr.grad, rtol=2e-3,
+                                       atol=2e-3)
