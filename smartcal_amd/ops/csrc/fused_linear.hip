@@ -130,20 +130,25 @@ extern "C" __global__ __launch_bounds__(512) void fused_linear_fwd_kernel(
         write_wtile(wb);
         if (t + 1 < NT_MAX) load_wtile(t + 1, wa);
       }
-      // split accumulator (even/odd k4): halves the dependent-MFMA chain
+      // front-load ALL LDS operand reads, then run the MFMA chain with a
+      // split accumulator: interleaved read/MFMA makes the compiler emit
+      // an lgkmcnt(0) wait per MFMA pair, exposing LDS latency each time
+      float ar[BK / 4], br[BK / 4];
+#pragma unroll
+      for (int k4 = 0; k4 < BK / 4; ++k4) {
+        const int k = k4 * 4 + l4;
+        ar[k4] = xs[l15 * (BK + 1) + k];
+        br[k4] = wsw[l15 * (BK + 1) + k];
+      }
+      // MFMA is a whole-wave op: never predicate it per-lane; edge
+      // tiles contribute zeros through the zero-padded LDS operands.
       f32x4 acc2 = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int k4 = 0; k4 < BK / 4; k4 += 2) {
-        // MFMA is a whole-wave op: never predicate it per-lane; edge
-        // tiles contribute zeros through the zero-padded LDS operands.
-        const int k0 = k4 * 4 + l4;
-        const int k1 = (k4 + 1) * 4 + l4;
-        float a0 = xs[l15 * (BK + 1) + k0];
-        float b0 = wsw[l15 * (BK + 1) + k0];
-        float a1 = xs[l15 * (BK + 1) + k1];
-        float b1 = wsw[l15 * (BK + 1) + k1];
-        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[t], 0, 0, 0);
-        acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc2, 0, 0, 0);
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(ar[k4], br[k4],
+                                                      acc[t], 0, 0, 0);
+        acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(ar[k4 + 1], br[k4 + 1],
+                                                    acc2, 0, 0, 0);
       }
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[t][r] += acc2[r];
@@ -423,18 +428,20 @@ extern "C" __global__ __launch_bounds__(1024) void mlp_chain_fwd_kernel(
           write_wtile(wb);
           if (t + 1 < CNT) load_wtile(t + 1, wa);
         }
+        // front-load all LDS operand reads (see single-layer kernel note)
+        float ar[BK / 4], br[BK / 4];
+#pragma unroll
+        for (int k4 = 0; k4 < BK / 4; ++k4) {
+          ar[k4] = xb[l15 * XP + kk + k4 * 4 + l4];
+          br[k4] = wsw[l15 * (BK + 1) + k4 * 4 + l4];
+        }
         f32x4 acc2 = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int k4 = 0; k4 < BK / 4; k4 += 2) {
-          const int k0 = kk + k4 * 4 + l4;
-          const int k1 = kk + (k4 + 1) * 4 + l4;
-          float a0 = xb[l15 * XP + k0];
-          float b0 = wsw[l15 * (BK + 1) + k4 * 4 + l4];
-          float a1 = xb[l15 * XP + k1];
-          float b1 = wsw[l15 * (BK + 1) + (k4 + 1) * 4 + l4];
-          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[t],
-                                                        0, 0, 0);
-          acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc2,
+          acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(ar[k4], br[k4],
+                                                        acc[t], 0, 0, 0);
+          acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(ar[k4 + 1],
+                                                      br[k4 + 1], acc2,
                                                       0, 0, 0);
         }
 #pragma unroll

@@ -23,27 +23,56 @@ extern "C" __global__ __launch_bounds__(256) void gemm_f32_nn_kernel(
   // NOTE: no early return — all waves must reach the barriers below;
   // out-of-range waves compute on zeros and skip their stores.
 
-  // A tile staged in LDS (A fragment reads are row-scattered in global);
-  // B fragment reads are naturally coalesced (consecutive lanes ->
-  // consecutive columns), so B streams from L2.
-  __shared__ float as[16 * 68];  // [16][64+4] pad 4: conflict-free b32 reads
+  // A tiles double-buffered in LDS (A fragment reads are row-scattered in
+  // global): the global loads of chunk i+1 are issued into registers
+  // before chunk i's MFMAs, so their latency hides under the matrix-core
+  // work — one barrier per chunk instead of two. B fragment reads are
+  // naturally coalesced (consecutive lanes -> consecutive columns), so B
+  // streams from L2.
+  __shared__ float as[2][16 * 68];  // [16][64+4] pad 4: conflict-free reads
   const int col = col0 + l15;   // B fragment col
+  // each thread stages 4 A elements per chunk (16*64/256)
+  const int st_r = (tid * 4) >> 6, st_c = (tid * 4) & 63;
+  auto load_a = [&](int kk, float4* v) {
+    const int gr = row0 + st_r;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int gc = kk + st_c + j;
+      (&v->x)[j] = (gr < M && gc < K) ? A[(long)gr * K + gc] : 0.f;
+    }
+  };
+  auto store_a = [&](int buf, const float4* v) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      as[buf][st_r * 68 + st_c + j] = (&v->x)[j];
+  };
   f32x4 acc = (f32x4){0.f, 0.f, 0.f, 0.f};
+  float4 areg;
+  load_a(0, &areg);
+  store_a(0, &areg);
+  __syncthreads();
+  int cur = 0;
   for (int kk = 0; kk < K; kk += 64) {
     const int kmax = min(64, K - kk);
-    for (int idx = tid; idx < 16 * 64; idx += 256) {
-      const int r = idx >> 6, c = idx & 63;
-      const int gr = row0 + r, gc = kk + c;
-      as[r * 68 + c] = (gr < M && gc < K) ? A[(long)gr * K + gc] : 0.f;
-    }
-    __syncthreads();
+    if (kk + 64 < K) load_a(kk + 64, &areg);  // prefetch next chunk
+    // issue ALL 16 B loads before the MFMA chain: interleaving them
+    // makes the compiler emit waitcnt vmcnt(0) before every MFMA, which
+    // exposes a full L2 round trip per instruction (~24 us/kernel);
+    // front-loading brings the chunk down to one latency + the MFMAs
+    float breg[16];
 #pragma unroll
     for (int k4 = 0; k4 < 16; ++k4) {
       const int k = k4 * 4 + l4;
-      const float a = as[l15 * 68 + k];
-      const float b = (col < N && k < kmax) ? B[(long)(kk + k) * N + col]
-                                            : 0.f;
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+      breg[k4] = (col < N && k < kmax) ? B[(long)(kk + k) * N + col] : 0.f;
+    }
+#pragma unroll
+    for (int k4 = 0; k4 < 16; ++k4) {
+      const float a = as[cur][l15 * 68 + k4 * 4 + l4];
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, breg[k4], acc, 0, 0, 0);
+    }
+    if (kk + 64 < K) {
+      store_a(1 - cur, &areg);
+      cur = 1 - cur;
     }
     __syncthreads();
   }
@@ -73,11 +102,20 @@ extern "C" __global__ __launch_bounds__(256) void gemm_f32_tn_kernel(
   const int arow = row0 + l15;  // output row = column of A
   const int bcol = col0 + l15;
   f32x4 acc = (f32x4){0.f, 0.f, 0.f, 0.f};
-  for (int kk = 0; kk < K; kk += 4) {
-    const int k = kk + l4;
-    const float a = (arow < M && k < K) ? A[(long)k * M + arow] : 0.f;
-    const float b = (bcol < N && k < K) ? B[(long)k * N + bcol] : 0.f;
-    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  // front-load operands in chunks of 8 k-steps so the MFMAs don't stall
+  // on a vmcnt(0) per instruction (see gemm_nn note)
+  for (int kk = 0; kk < K; kk += 32) {
+    float areg[8], breg[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = kk + j * 4 + l4;
+      areg[j] = (arow < M && k < K) ? A[(long)k * M + arow] : 0.f;
+      breg[j] = (bcol < N && k < K) ? B[(long)k * N + bcol] : 0.f;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(areg[j], breg[j], acc,
+                                                 0, 0, 0);
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
