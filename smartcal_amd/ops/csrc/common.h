@@ -8,29 +8,47 @@
 
 #define WAVE 64
 
+// DPP lane permute within a 16-lane row: quad_perm XOR-1 / XOR-2, then
+// row-rotate by 4 and 8. DPP runs in the VALU pipe (a few cycles) where
+// __shfl_xor lowers to ds_bpermute through the LDS pipe (~10x the
+// latency, plus an lgkmcnt wait per step).
+template <int CTRL>
+__device__ __forceinline__ float _dpp_f(float v) {
+  union {
+    float f;
+    int i;
+  } x;
+  x.f = v;
+  x.i = __builtin_amdgcn_update_dpp(0, x.i, CTRL, 0xF, 0xF, true);
+  return x.f;
+}
+
+// Sum across a 16-lane group (lanes sharing lane>>4); all 16 get the
+// total. 4 DPP ops, no LDS traffic.
+__device__ __forceinline__ float group16_sum(float v) {
+  v += _dpp_f<0xB1>(v);     // quad_perm [1,0,3,2]  (xor 1)
+  v += _dpp_f<0x4E>(v);     // quad_perm [2,3,0,1]  (xor 2)
+  v += _dpp_f<0x124>(v);    // row_ror:4
+  v += _dpp_f<0x128>(v);    // row_ror:8
+  return v;
+}
+
 // Sum across all 64 lanes of a wave; every lane gets the total.
+// 4 DPP ops for the in-row sum + 2 cross-row butterflies.
 __device__ __forceinline__ float wave_sum(float v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) {
-    v += __shfl_xor(v, off, WAVE);
-  }
+  v = group16_sum(v);
+  v += __shfl_xor(v, 16, WAVE);
+  v += __shfl_xor(v, 32, WAVE);
   return v;
 }
 
 __device__ __forceinline__ float wave_max(float v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) {
-    v = fmaxf(v, __shfl_xor(v, off, WAVE));
-  }
-  return v;
-}
-
-// Sum across a 16-lane group (lanes sharing lane>>4); all 16 get the total.
-__device__ __forceinline__ float group16_sum(float v) {
-#pragma unroll
-  for (int off = 8; off > 0; off >>= 1) {
-    v += __shfl_xor(v, off, WAVE);
-  }
+  v = fmaxf(v, _dpp_f<0xB1>(v));
+  v = fmaxf(v, _dpp_f<0x4E>(v));
+  v = fmaxf(v, _dpp_f<0x124>(v));
+  v = fmaxf(v, _dpp_f<0x128>(v));
+  v = fmaxf(v, __shfl_xor(v, 16, WAVE));
+  v = fmaxf(v, __shfl_xor(v, 32, WAVE));
   return v;
 }
 

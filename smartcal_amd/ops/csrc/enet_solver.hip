@@ -44,41 +44,53 @@ struct LdsLayout {
   float* __restrict__ xprev;  // M (previous-epoch x snapshot)
 };
 
+// Per-lane register cache of A: row[j] = A[lane][j] (for the forward
+// matvec, lanes < N) and col[i] = A[i][lane] (for the A^T r gradient,
+// lanes < M). A is constant for the whole solve; keeping it in VGPRs
+// and front-loading x/r turns each closure evaluation into two LDS
+// waits + pure VALU FMA chains (the interleaved LDS-read form stalled
+// on an lgkmcnt wait per FMA).
+struct RegA {
+  float row[32];
+  float col[32];
+};
+
 // residual + loss + gradient of ||y-Ax||^2 + rho1||x||^2 + rho2||x||_1
-__device__ static float eval_loss_grad(const LdsLayout& L, int N, int M,
-                                       float rho1, float rho2) {
+__device__ __forceinline__ static float eval_loss_grad(
+    const LdsLayout& L, const RegA& Ar, int N, int M, float rho1,
+    float rho2) {
   const int lane = threadIdx.x;
-  float ri = 0.f;
-  if (lane < N) {
-    // 4 independent accumulators: breaks the dependent FMA/LDS chain
-    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-    int j = 0;
-    for (; j + 3 < M; j += 4) {
-      a0 += L.A[lane * M + j + 0] * L.x[j + 0];
-      a1 += L.A[lane * M + j + 1] * L.x[j + 1];
-      a2 += L.A[lane * M + j + 2] * L.x[j + 2];
-      a3 += L.A[lane * M + j + 3] * L.x[j + 3];
-    }
-    for (; j < M; ++j) a0 += L.A[lane * M + j] * L.x[j];
-    ri = L.y[lane] - ((a0 + a1) + (a2 + a3));
-    L.r[lane] = ri;
+  float xr[32];
+#pragma unroll
+  for (int j = 0; j < 32; ++j) xr[j] = (j < M) ? L.x[j] : 0.f;
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+#pragma unroll
+  for (int j = 0; j < 32; j += 4) {
+    a0 += Ar.row[j + 0] * xr[j + 0];
+    a1 += Ar.row[j + 1] * xr[j + 1];
+    a2 += Ar.row[j + 2] * xr[j + 2];
+    a3 += Ar.row[j + 3] * xr[j + 3];
   }
-  float xl = (lane < M) ? L.x[lane] : 0.f;
+  const float ri = (lane < N) ? L.y[lane] - ((a0 + a1) + (a2 + a3)) : 0.f;
+  if (lane < N) L.r[lane] = ri;
+  const float xl = (lane < M) ? L.x[lane] : 0.f;
   float loss = wave_sum(ri * ri + rho1 * xl * xl + rho2 * fabsf(xl));
   __builtin_amdgcn_s_barrier();  // r[] visible (single wave: lockstep, but
                                  // keep an explicit ordering point)
+  float rr[32];
+#pragma unroll
+  for (int i = 0; i < 32; ++i) rr[i] = (i < N) ? L.r[i] : 0.f;
+  float b0 = 0.f, b1 = 0.f, b2 = 0.f, b3 = 0.f;
+#pragma unroll
+  for (int i = 0; i < 32; i += 4) {
+    b0 += Ar.col[i + 0] * rr[i + 0];
+    b1 += Ar.col[i + 1] * rr[i + 1];
+    b2 += Ar.col[i + 2] * rr[i + 2];
+    b3 += Ar.col[i + 3] * rr[i + 3];
+  }
   if (lane < M) {
-    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-    int i = 0;
-    for (; i + 3 < N; i += 4) {
-      a0 += L.A[(i + 0) * M + lane] * L.r[i + 0];
-      a1 += L.A[(i + 1) * M + lane] * L.r[i + 1];
-      a2 += L.A[(i + 2) * M + lane] * L.r[i + 2];
-      a3 += L.A[(i + 3) * M + lane] * L.r[i + 3];
-    }
-    for (; i < N; ++i) a0 += L.A[i * M + lane] * L.r[i];
-    float atr = (a0 + a1) + (a2 + a3);
-    float sgn = (xl > 0.f) ? 1.f : (xl < 0.f ? -1.f : 0.f);
+    const float atr = (b0 + b1) + (b2 + b3);
+    const float sgn = (xl > 0.f) ? 1.f : (xl < 0.f ? -1.f : 0.f);
     L.g[lane] = -2.f * atr + 2.f * rho1 * xl + rho2 * sgn;
   }
   return loss;
@@ -97,12 +109,13 @@ __device__ static float lds_absmax(const float* a, int n) {
 }
 
 // phi(t) = f(x0 + t d); leaves x at the evaluated point, g = grad there.
-__device__ static float ls_eval(const LdsLayout& L, int N, int M,
+__device__ static float ls_eval(const LdsLayout& L, const RegA& Ar,
+                                int N, int M,
                                 float rho1, float rho2, float t,
                                 float* gtd_out) {
   const int lane = threadIdx.x;
   if (lane < M) L.x[lane] = L.x0[lane] + t * L.d[lane];
-  float f = eval_loss_grad(L, N, M, rho1, rho2);
+  float f = eval_loss_grad(L, Ar, N, M, rho1, rho2);
   *gtd_out = lds_dot(L.g, L.d, M);
   return f;
 }
@@ -126,7 +139,8 @@ __device__ static float cubic_interp(float x1, float f1, float g1, float x2,
 // Strong-Wolfe line search (same algorithm as optim/lbfgs.py::_strong_wolfe).
 // Returns f at the accepted point; x and g hold that point on exit; *t_io
 // the accepted step.
-__device__ static float strong_wolfe(const LdsLayout& L, int N, int M,
+__device__ static float strong_wolfe(const LdsLayout& L, const RegA& Ar,
+                                     int N, int M,
                                      float rho1, float rho2, float f0,
                                      float gtd0, float* t_io) {
   const int lane = threadIdx.x;
@@ -148,7 +162,7 @@ __device__ static float strong_wolfe(const LdsLayout& L, int N, int M,
   float f_new, gtd_new;
   bool bracketed = false;
   while (ls_iter < max_ls) {
-    f_new = ls_eval(L, N, M, rho1, rho2, t, &gtd_new);
+    f_new = ls_eval(L, Ar, N, M, rho1, rho2, t, &gtd_new);
     if (f_new > (f0 + c1 * t * gtd0) || (ls_iter > 0 && f_new >= f_prev)) {
       br_t[0] = t_prev; br_f[0] = f_prev; br_gtd[0] = gtd_prev;
       // bg0 already holds g(t_prev)
@@ -208,7 +222,7 @@ __device__ static float strong_wolfe(const LdsLayout& L, int N, int M,
     } else {
       insuf = false;
     }
-    f_new = ls_eval(L, N, M, rho1, rho2, t, &gtd_new);
+    f_new = ls_eval(L, Ar, N, M, rho1, rho2, t, &gtd_new);
     if (f_new > (f0 + c1 * t * gtd0) || f_new >= br_f[low]) {
       br_t[high] = t; br_f[high] = f_new; br_gtd[high] = gtd_new;
       if (lane < M) br_g[high][lane] = L.g[lane];
@@ -274,6 +288,15 @@ extern "C" __global__ __launch_bounds__(64) void enet_lbfgs_solve_kernel(
   L.xprev = p; p += M;
 
   for (int i = lane; i < N * M; i += WAVE) L.A[i] = Ag[(long)env * N * M + i];
+  __builtin_amdgcn_s_barrier();
+  // per-lane register cache of A (row for the matvec, column for A^T r)
+  RegA Ar;
+#pragma unroll
+  for (int j = 0; j < 32; ++j)
+    Ar.row[j] = (lane < N && j < M) ? L.A[lane * M + j] : 0.f;
+#pragma unroll
+  for (int i = 0; i < 32; ++i)
+    Ar.col[i] = (lane < M && i < N) ? L.A[i * M + lane] : 0.f;
   if (lane < N) L.y[lane] = yg[(long)env * N + lane];
   if (lane < M) {
     L.x[lane] = 0.f;
@@ -290,7 +313,7 @@ extern "C" __global__ __launch_bounds__(64) void enet_lbfgs_solve_kernel(
   if (lane < M) L.xprev[lane] = 1e30f;
 
   for (int epoch = 0; epoch < epochs; ++epoch) {
-    float loss = eval_loss_grad(L, N, M, rho1, rho2);
+    float loss = eval_loss_grad(L, Ar, N, M, rho1, rho2);
     // epoch-level early stop: once the whole optimizer has converged the
     // reference just keeps burning closure evaluations; identical x at
     // fp32, ~3-5x fewer evaluations.
@@ -369,7 +392,7 @@ extern "C" __global__ __launch_bounds__(64) void enet_lbfgs_solve_kernel(
       }
 
       if (lane < M) L.x0[lane] = L.x[lane];
-      loss = strong_wolfe(L, N, M, rho1, rho2, loss, gtd, &t);
+      loss = strong_wolfe(L, Ar, N, M, rho1, rho2, loss, gtd, &t);
 
       if (lds_absmax(L.g, M) <= TOL_GRAD) { outer_done = true; break; }
       float step_max = lds_absmax(L.d, M) * fabsf(t);
