@@ -88,3 +88,21 @@ def shapelet_basis(n0: int, beta: float, l: np.ndarray,
                                  * math.sqrt(math.pi) * beta)
             out[a * n0 + b] = ha * _hermite(b, xm) * gm * nb
     return out
+
+
+def correct_shapelet_modes(text: str) -> str:
+    """Rescale shapelet coefficients by i!/(i+1)! per row block — parity
+    with `calibration/correct_shapelet_modes.py:4-30` (mode-file format
+    conversion between SAGECal versions)."""
+    lines = text.splitlines()
+    out = [lines[0], lines[1]]
+    n0 = int(lines[1].split()[0])
+    idx = 2
+    for ci in range(n0):
+        scale = 1.0 / (ci + 1)          # i!/(i+1)!
+        for cj in range(n0):
+            num, val = lines[idx].split()
+            out.append(f"{num} {float(val) * scale}")
+            idx += 1
+    out.extend(lines[idx:])
+    return "\n".join(out) + "\n"

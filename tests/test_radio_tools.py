@@ -201,3 +201,37 @@ def test_shapelet_roundtrip():
     basis = shapelet.shapelet_basis(3, 0.1, l, np.zeros_like(l))
     assert basis.shape == (9, 101)
     assert np.argmax(basis[0]) == 50
+
+
+def test_vis_io_roundtrip(tmp_path):
+    from smartcal_amd.radio import io as rio
+    from smartcal_amd.radio import array as arr, sim
+    rng = np.random.default_rng(4)
+    layout = arr.lofar_like_layout(N=5, rng=rng)
+    skym, cs, *_, ra0, dec0 = sim.make_demixing_sky(rng, n_outliers=1)
+    vis = sim.simulate_observation(layout, skym, cs,
+                                   np.array([150e6]), ra0, dec0,
+                                   Ts=1, Tdelta=2, snr=10, rng=rng,
+                                   torch_seed=0)
+    text = rio.write_corr_text(vis.uvw, vis.data[0])
+    uvw2, v2 = rio.read_corr_text(text)
+    np.testing.assert_allclose(uvw2.numpy(), vis.uvw.numpy(), rtol=1e-6)
+    np.testing.assert_allclose(v2.numpy(), vis.data[0].numpy(), rtol=1e-5,
+                               atol=1e-6)
+    p = str(tmp_path / "vis.npz")
+    rio.save_visdata(vis, p)
+    vis2 = rio.load_visdata(p)
+    np.testing.assert_allclose(vis2.data.numpy(), vis.data.numpy())
+    assert vis2.N == vis.N and vis2.Tdelta == vis.Tdelta
+
+
+def test_correct_shapelet_modes():
+    from smartcal_amd.radio import shapelet
+    rng = np.random.default_rng(1)
+    text, n0, beta, coeff = shapelet.generate_random_shapelet_model(rng)
+    out = shapelet.correct_shapelet_modes(text)
+    _, n0b, betab, coeffb = shapelet.parse_shapelet_model(out)
+    assert n0b == n0
+    # row block ci scaled by 1/(ci+1)
+    want = coeff.reshape(n0, n0) / (np.arange(1, n0 + 1)[:, None])
+    np.testing.assert_allclose(coeffb.reshape(n0, n0), want, rtol=1e-6)
