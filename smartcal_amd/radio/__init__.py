@@ -13,4 +13,5 @@ compatibility (`smartcal_amd.radio.sky`, `smartcal_amd.radio.solutions`).
 """
 
 from . import (coords, sky, coherency, consensus, hessian, solutions,  # noqa: F401
-               array, sim, solver, imaging, influence)  # noqa: F401
+               array, sim, solver, imaging, influence, shapelet,  # noqa: F401
+               dataset, io, small_complex)  # noqa: F401
