@@ -105,3 +105,31 @@ def test_cnn_sac_gpu_learn():
                                hint=np.zeros(6, np.float32))
     agent.learn()
     assert torch.isfinite(agent.actor_fp.flat).all()
+
+
+@needs_gpu
+def test_training_example_gpu():
+    from smartcal_amd.radio.dataset import generate_training_example
+    rng = np.random.default_rng(3)
+    x, y, K = generate_training_example(rng, Ninf=64, N_stations=24,
+                                        device="cuda")
+    assert x.shape == (K * (64 * 64 + 8),)
+    assert np.isfinite(x).all()
+    assert y.shape == (K - 1,)
+
+
+@needs_gpu
+def test_transformer_gpu():
+    from smartcal_amd.models import TransformerEncoder
+    torch.manual_seed(0)
+    K = 6
+    Nout = 64 * 64 + 8
+    net = TransformerEncoder(num_layers=1, input_dim=K * Nout,
+                             model_dim=K * 66, num_classes=K - 1,
+                             num_heads=K, dropout=0.0).cuda()
+    x = torch.randn(8, K * Nout, device="cuda")
+    y = net(x)
+    loss = torch.nn.functional.binary_cross_entropy(
+        y, torch.rand(8, K - 1, device="cuda"))
+    loss.backward()
+    assert torch.isfinite(y).all()
