@@ -86,7 +86,8 @@ class Actor:
     epochs×steps, uploads its transition block."""
 
     def __init__(self, env, agent, codec: _Codec, max_transitions: int,
-                 epochs: int = 10, steps: int = 10, use_hint: bool = False):
+                 epochs: int = 10, steps: int = 10, use_hint: bool = False,
+                 env_factory=None):
         self.env = env
         self.agent = agent
         self.codec = codec
@@ -94,32 +95,46 @@ class Actor:
         self.epochs = epochs
         self.steps = steps
         self.use_hint = use_hint
+        self.env_factory = env_factory
 
     def run_round(self):
         # 1. receive current actor weights (flat broadcast from rank 0)
         dist.broadcast(self.agent.actor_fp.flat, src=0)
-        # 2. local rollouts
+        # 2. local rollouts — failure-tolerant: an env error (numerical
+        # blow-up, bad scenario draw) must not wedge the collective
+        # schedule, so the actor reports what it has (possibly 0
+        # transitions), rebuilds its env, and stays in the ring. (The
+        # reference has no failure handling at all — SURVEY.md §5.)
         block = torch.zeros((self.cap, self.codec.width),
                             dtype=torch.float32)
         count = 0
-        for _ in range(self.epochs):
-            obs = self.env.reset()
-            hint = getattr(self.env, "hint", None) if self.use_hint else None
-            for _ in range(self.steps):
-                action = self.agent.choose_action(obs)
-                out = self.env.step(action)
-                if len(out) == 5:
-                    obs_, reward, done, hint, _ = out
-                else:
-                    obs_, reward, done, _ = out
-                if count < self.cap:
-                    block[count] = self.codec.pack(obs, action, reward,
-                                                   obs_, done, hint)
-                    count += 1
-                obs = obs_
-                if done:
-                    break
-        block[0, -1] = block[0, -1]  # no-op; count goes in a header tensor
+        try:
+            for _ in range(self.epochs):
+                obs = self.env.reset()
+                hint = getattr(self.env, "hint", None) if self.use_hint \
+                    else None
+                for _ in range(self.steps):
+                    action = self.agent.choose_action(obs)
+                    out = self.env.step(action)
+                    if len(out) == 5:
+                        obs_, reward, done, hint, _ = out
+                    else:
+                        obs_, reward, done, _ = out
+                    if count < self.cap:
+                        block[count] = self.codec.pack(obs, action, reward,
+                                                       obs_, done, hint)
+                        count += 1
+                    obs = obs_
+                    if done:
+                        break
+        except Exception as e:  # noqa: BLE001
+            print(f"[actor] env failure ({type(e).__name__}: {e}); "
+                  f"uploading {count} transitions and rebuilding env")
+            if self.env_factory is not None:
+                try:
+                    self.env = self.env_factory()
+                except Exception:  # noqa: BLE001
+                    pass
         hdr = torch.tensor([float(count)])
         # 3. upload to the learner
         dist.gather(hdr, dst=0)
@@ -198,7 +213,8 @@ def run_process(rank: int, world_size: int, agent_factory, env_factory,
         agent = agent_factory()
         env = env_factory()
         actor = Actor(env, agent, codec, max_transitions, epochs=epochs,
-                      steps=steps, use_hint=use_hint)
+                      steps=steps, use_hint=use_hint,
+                      env_factory=env_factory)
         for _ in range(episodes):
             actor.run_round()
         return None

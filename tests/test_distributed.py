@@ -162,3 +162,59 @@ def test_solver_freq_sharded_consensus():
     # and each rank's local residual is well below its data power
     for r in range(2):
         assert out[r][1] < 0.2 * out[r][2]
+
+
+def _fault_worker(rank, world, port, q):
+    from smartcal_amd.distributed.learner_actor import (run_process,
+                                                        Actor, _Codec)
+    from smartcal_amd.envs.enet import ENetEnv
+    from smartcal_amd.rl.sac import Agent
+
+    torch.manual_seed(rank)
+    np.random.seed(rank)
+
+    class FlakyEnv(ENetEnv):
+        """Env that blows up on its second reset (then works again)."""
+        calls = 0
+
+        def reset(self):
+            FlakyEnv.calls += 1
+            if FlakyEnv.calls == 2:
+                raise RuntimeError("synthetic env failure")
+            return super().reset()
+
+    def agent_factory():
+        return Agent(gamma=0.99, batch_size=4, n_actions=NACT, tau=0.005,
+                     max_mem_size=64, input_dims=[OBS_DIM], lr_a=1e-3,
+                     lr_c=1e-3, reward_scale=N, alpha=0.03,
+                     device=torch.device("cpu"))
+
+    def env_factory():
+        return FlakyEnv(M, N, device=torch.device("cpu"))
+
+    scores = run_process(rank, world, agent_factory, env_factory,
+                         obs_dim=OBS_DIM, n_actions=NACT, episodes=3,
+                         epochs=1, steps=2, learner_addr="127.0.0.1",
+                         learner_port=port, max_transitions=8,
+                         backend="gloo")
+    if rank == 0:
+        q.put(scores)
+
+
+@pytest.mark.timeout(300)
+def test_learner_tolerates_actor_env_failure():
+    """A mid-round env crash must not wedge the collective schedule:
+    the actor uploads a short round and keeps going."""
+    port = 29561
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_fault_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+    scores = q.get()
+    # 3 episodes ran; the failed round contributed no transitions
+    assert len(scores) >= 2
