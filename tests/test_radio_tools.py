@@ -235,3 +235,24 @@ def test_correct_shapelet_modes():
     # row block ci scaled by 1/(ci+1)
     want = coeff.reshape(n0, n0) / (np.arange(1, n0 + 1)[:, None])
     np.testing.assert_allclose(coeffb.reshape(n0, n0), want, rtol=1e-6)
+
+
+def test_convert_model_bbs():
+    from smartcal_amd.radio.convert import convert_model, parse_bbs_text
+    bbs = """# (Name, Type, Patch, Ra, Dec, I, Q, U, V, ReferenceFrequency, SpectralIndex, MajorAxis, MinorAxis, Orientation) = format
+s1, POINT, CasA, 23:23:24.0, 58.48.54.0, 100.0, 0, 0, 0, 150e6, [-0.7], , ,
+s2, GAUSSIAN, CasA, 23:23:30.0, 58.49.00.0, 50.0, 0, 0, 0, 150e6, [-0.5], 60, 30, 45
+t1, POINT, Target, 12:00:00.0, 45.00.00.0, 2.0, 0, 0, 0, 150e6, [0.1], , ,
+"""
+    skym, clusters = parse_bbs_text(bbs)
+    assert len(skym) == 3
+    assert len(clusters) == 2
+    assert skym.gaussian.sum() == 1
+    assert abs(skym.ra[0] - (23 + 23 / 60 + 24 / 3600) * np.pi / 12) < 1e-9
+    sky_t, cl_t, rho_t = convert_model(bbs, start_cluster=2)
+    assert "GCasA1" in sky_t
+    from smartcal_amd.radio.sky import parse_sky_text, parse_cluster_text
+    sky2 = parse_sky_text(sky_t)
+    assert len(sky2) == 3
+    cl2 = parse_cluster_text(cl_t)
+    assert cl2[0].cid == 2
