@@ -521,7 +521,7 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
     for (int pi = 0; pi < N - 1; ++pi)
       if (lane < N && lane > pi) off += B[pi * N + lane] * B[pi * N + lane];
     off = wave_sum(off);
-    if (off < 1e-12f * dscale) break;
+    if (off < 1e-11f * dscale) break;
     for (int pi = 0; pi < N - 1; ++pi) {
       for (int q = pi + 1; q < N; ++q) {
         float apq = B[pi * N + q];

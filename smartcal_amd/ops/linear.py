@@ -34,6 +34,21 @@ _ACT_CODES = {"none": ACT_NONE, "elu": ACT_ELU, "relu": ACT_RELU,
               "tanh": ACT_TANH}
 
 
+_scratch: dict = {}
+
+
+def _scratch_buf(n: int, device) -> torch.Tensor:
+    """Reusable throwaway accumulation buffer for dgamma/dbeta when the
+    parameter grads are not wanted (frozen-critic actor phase): avoids
+    two at::zeros fill kernels per layer per backward."""
+    key = (device.type, device.index)
+    buf = _scratch.get(key)
+    if buf is None or buf.numel() < n:
+        buf = torch.empty(max(n, 1024), device=device)
+        _scratch[key] = buf
+    return buf[:n]
+
+
 def _grad_view(p: torch.Tensor) -> torch.Tensor:
     """The tensor the kernels accumulate this parameter's gradient into.
 
@@ -71,12 +86,17 @@ class _FusedLinearFn(torch.autograd.Function):
             dz = ext().fused_linear_bwd_dz_into(
                 dy, y, zhat, rstd, gamma, ctx.act_code, True,
                 _grad_view(gp), _grad_view(bep))
+        elif ctx.with_ln:
+            # grads not wanted (frozen params): accumulate into a
+            # reusable scratch buffer instead of fresh zeros
+            n = dy.shape[1]
+            sb = _scratch_buf(2 * n, dy.device)
+            dz = ext().fused_linear_bwd_dz_into(
+                dy, y, zhat, rstd, gamma, ctx.act_code, True,
+                sb[:n], sb[n:2 * n])
         else:
-            dz, dgamma, dbeta = ext().fused_linear_bwd_dz(
-                dy, y, zhat, rstd, gamma, ctx.act_code, ctx.with_ln)
-            if ctx.with_ln and want_w:
-                _grad_view(gp).add_(dgamma)
-                _grad_view(bep).add_(dbeta)
+            dz, _, _ = ext().fused_linear_bwd_dz(
+                dy, y, zhat, rstd, gamma, ctx.act_code, False)
         dx = ext().mfma_gemm_nn(dz, W)          # (B,N) @ (N,K) -> (B,K)
         if want_w:
             # dW = dz^T @ x and db = col-sum(dz), accumulated straight
@@ -190,8 +210,11 @@ class _FusedChainFn(torch.autograd.Function):
                     dy, ys[l], zhats[l], rstds[l], g, ctx.acts[l], True,
                     _grad_view(g), _grad_view(be))
             else:
-                dz, _, _ = ext().fused_linear_bwd_dz(
-                    dy, ys[l], zhats[l], rstds[l], g, ctx.acts[l], True)
+                n = dy.shape[1]
+                sb = _scratch_buf(2 * n, dy.device)
+                dz = ext().fused_linear_bwd_dz_into(
+                    dy, ys[l], zhats[l], rstds[l], g, ctx.acts[l], True,
+                    sb[:n], sb[n:2 * n])
             x_l = x if l == 0 else ys[l - 1]
             if want_w:
                 ext().mfma_gemm_tn_bias_into(dz, x_l, _grad_view(W),
