@@ -87,9 +87,19 @@ class ReplayBuffer:
     # -- checkpointing (reference pickles the whole object,
     #    ``enet_sac.py:59-73``; we save a state dict of host tensors) ------
     def save_checkpoint(self, filename: Optional[str] = None):
+        # only the filled prefix is stored (empty slots are zeros)
         fn = filename or self.filename
-        sd = {k: (v.cpu() if torch.is_tensor(v) else v)
-              for k, v in self.__dict__.items() if k != "device"}
+        n = len(self)
+        sd = {}
+        for k, v in self.__dict__.items():
+            if k == "device":
+                continue
+            if torch.is_tensor(v) and v.shape[:1] == (self.mem_size,):
+                sd[k] = v[:n].cpu().clone()
+            elif torch.is_tensor(v):
+                sd[k] = v.cpu()
+            else:
+                sd[k] = v
         with open(fn, "wb") as f:
             pickle.dump(sd, f)
 
@@ -98,8 +108,12 @@ class ReplayBuffer:
         with open(fn, "rb") as f:
             sd = pickle.load(f)
         for k, v in sd.items():
-            if torch.is_tensor(v) and torch.is_tensor(getattr(self, k, None)):
-                getattr(self, k).copy_(v.to(self.device))
+            cur = getattr(self, k, None)
+            if torch.is_tensor(v) and torch.is_tensor(cur):
+                if v.shape == cur.shape:
+                    cur.copy_(v.to(self.device))
+                else:       # filled-prefix checkpoint
+                    cur[:v.shape[0]].copy_(v.to(self.device))
             else:
                 setattr(self, k, v)
 

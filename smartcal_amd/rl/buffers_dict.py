@@ -84,9 +84,20 @@ class DictReplayBuffer:
                 self.terminal_memory[idx], self.hint_memory[idx])
 
     def save_checkpoint(self, filename=None):
+        # store only the filled prefix: a fresh 16000-slot image buffer
+        # is ~1.3 GB of zeros otherwise
         fn = filename or self.filename
-        sd = {k: (v.cpu() if torch.is_tensor(v) else v)
-              for k, v in self.__dict__.items() if k != "device"}
+        n = len(self)
+        sd = {}
+        for k, v in self.__dict__.items():
+            if k == "device":
+                continue
+            if torch.is_tensor(v) and v.shape[:1] == (self.mem_size,):
+                sd[k] = v[:n].cpu().clone()
+            elif torch.is_tensor(v):
+                sd[k] = v.cpu()
+            else:
+                sd[k] = v
         with open(fn, "wb") as f:
             pickle.dump(sd, f)
 
@@ -95,8 +106,12 @@ class DictReplayBuffer:
         with open(fn, "rb") as f:
             sd = pickle.load(f)
         for k, v in sd.items():
-            if torch.is_tensor(v) and torch.is_tensor(getattr(self, k, None)):
-                getattr(self, k).copy_(v.to(self.device))
+            cur = getattr(self, k, None)
+            if torch.is_tensor(v) and torch.is_tensor(cur):
+                if v.shape == cur.shape:
+                    cur.copy_(v.to(self.device))
+                else:       # filled-prefix checkpoint
+                    cur[:v.shape[0]].copy_(v.to(self.device))
             else:
                 setattr(self, k, v)
 
