@@ -22,6 +22,7 @@ from torch.distributions import Normal
 
 from ..utils.device import default_device
 from ..utils.flatten import FlatParams, FusedAdam
+from ..utils.streams import StreamFork
 from .buffers_dict import DictPERBuffer, DictReplayBuffer
 from .conv_networks import CriticCNN, SACActorCNN, EPS, _init_layer
 
@@ -153,6 +154,7 @@ class Agent:
         self.rho = torch.tensor(0.0, device=self.device)
         self.admm_rho = admm_rho
         self.learn_counter = 0
+        self._fork = StreamFork(self.device)
         self._hard_sync()
 
     # ------------------------------------------------------------------
@@ -214,16 +216,25 @@ class Agent:
         with torch.no_grad():
             na, nlp = self.actor.sample_normal(img_, meta_,
                                                reparameterize=False)
-            q1_t = self.target_critic_1(img_, meta_, na)
-            q2_t = self.target_critic_2(img_, meta_, na)
+
+        def _t1():
+            with torch.no_grad():
+                return self.target_critic_1(img_, meta_, na)
+
+        def _t2():
+            with torch.no_grad():
+                return self.target_critic_2(img_, meta_, na)
+
+        # all four critic-family forwards overlap on 4 HIP streams
+        q1, q2, q1_t, q2_t = self._fork(
+            lambda: self.critic_1(img, meta, action),
+            lambda: self.critic_2(img, meta, action), _t1, _t2)
+        with torch.no_grad():
             tgt = torch.min(q1_t, q2_t) - self.alpha * nlp
             tgt = tgt.masked_fill(done, 0.0)
             # note: the CNN reference does NOT apply reward_scale in the
             # target (`calib_sac.py:345`); self.scale kept for parity only
             new_q = reward + self.gamma * tgt
-
-        q1 = self.critic_1(img, meta, action)
-        q2 = self.critic_2(img, meta, action)
         if is_w is not None:
             c_loss = (is_w * (q1 - new_q).pow(2)).mean() \
                 + (is_w * (q2 - new_q).pow(2)).mean()
@@ -241,8 +252,9 @@ class Agent:
 
         actions, log_probs = self.actor.sample_normal(img, meta,
                                                       reparameterize=True)
-        q1_pi = self.critic_1(img, meta, actions)
-        q2_pi = self.critic_2(img, meta, actions)
+        q1_pi, q2_pi = self._fork(
+            lambda: self.critic_1(img, meta, actions),
+            lambda: self.critic_2(img, meta, actions))
         critic_value = torch.min(q1_pi, q2_pi)
         actor_loss = (self.alpha * log_probs - critic_value).mean()
         if self.use_hint:

@@ -19,6 +19,7 @@ import torch.nn.functional as F
 
 from ..utils.device import default_device
 from ..utils.flatten import FlatParams, FusedAdam
+from ..utils.streams import StreamFork
 from .buffers_dict import DictPERBuffer, DictReplayBuffer
 from .conv_networks import CriticCNN, DeterministicActorCNN
 
@@ -76,6 +77,7 @@ class Agent:
         self.critic_2_opt = FusedAdam(self.critic_2_fp, lr=lr_c)
 
         self.use_hint = use_hint
+        self._fork = StreamFork(self.device)
         self.admm_rho = admm_rho
         self.Nadmm = 5
         self.adaptive_admm = True
@@ -140,13 +142,20 @@ class Agent:
             # target-policy smoothing (`enet_td3.py:247-251`)
             ta = ta + torch.clamp(torch.randn_like(ta) * 0.2, -0.5, 0.5)
             ta = torch.clamp(ta, self.min_action, self.max_action)
-            q1_t = self.target_critic_1(img_, meta_, ta)
-            q2_t = self.target_critic_2(img_, meta_, ta)
+        def _t1():
+            with torch.no_grad():
+                return self.target_critic_1(img_, meta_, ta)
+
+        def _t2():
+            with torch.no_grad():
+                return self.target_critic_2(img_, meta_, ta)
+
+        q1, q2, q1_t, q2_t = self._fork(
+            lambda: self.critic_1(img, meta, action),
+            lambda: self.critic_2(img, meta, action), _t1, _t2)
+        with torch.no_grad():
             q_t = torch.min(q1_t, q2_t).masked_fill(done, 0.0)
             target = reward + self.gamma * q_t
-
-        q1 = self.critic_1(img, meta, action)
-        q2 = self.critic_2(img, meta, action)
         if is_w is not None:
             c_loss = (is_w * (q1 - target).pow(2)).mean() \
                 + (is_w * (q2 - target).pow(2)).mean()

@@ -1,0 +1,50 @@
+"""Multi-stream fork/join helper for overlapping independent GPU work.
+
+Small RL networks run at a few-percent occupancy per kernel; running the
+twin critics / target critics concurrently on side HIP streams roughly
+halves their wall time. The event fork/join pattern is hipGraph-capture
+safe (captured as parallel graph branches) and stream-aware autograd
+runs the corresponding backwards concurrently too (see `rl/sac.py`).
+"""
+
+from __future__ import annotations
+
+import torch
+
+__all__ = ["StreamFork"]
+
+
+class StreamFork:
+    def __init__(self, device: torch.device):
+        self.device = device
+        self._streams: list = []
+        self._fork_evs: list = []
+        self._join_evs: list = []
+
+    def __call__(self, *fns):
+        """Run fns[0] on the current stream and each other fn on its own
+        side stream; returns their results as a tuple."""
+        if self.device.type != "cuda":
+            return tuple(f() for f in fns)
+        nside = len(fns) - 1
+        while len(self._streams) < nside:
+            self._streams.append(torch.cuda.Stream())
+            self._fork_evs.append(torch.cuda.Event())
+            self._join_evs.append(torch.cuda.Event())
+        results = [None] * len(fns)
+        for i in range(nside):
+            self._fork_evs[i].record()
+            with torch.cuda.stream(self._streams[i]):
+                self._fork_evs[i].wait()
+                results[i + 1] = fns[i + 1]()
+                self._join_evs[i].record()
+        results[0] = fns[0]()
+        for i in range(nside):
+            self._join_evs[i].wait()
+        if not torch.cuda.is_current_stream_capturing():
+            cs = torch.cuda.current_stream()
+            for r in results[1:]:
+                for t in (r if isinstance(r, (tuple, list)) else (r,)):
+                    if torch.is_tensor(t):
+                        t.record_stream(cs)
+        return tuple(results)
