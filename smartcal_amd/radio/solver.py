@@ -89,6 +89,17 @@ def _solve_sweeps(data22, C22, J, rho_t, prox_target, p_idx, q_idx, N,
     dev = data22.device
     t_int = (torch.arange(T, device=dev) // Tdelta).clamp_(max=Ts - 1)
     eyeK = torch.eye(2 * K, dtype=data22.dtype, device=dev)
+    # gather plan for the per-(interval, station) reduction: station n of
+    # interval ti receives exactly Tdelta·(N−1) contributions (as the p
+    # side or the q side of a baseline). Precomputing the gather indices
+    # into cat([p-side, q-side]) turns four slow complex index_add_
+    # scatters per sweep into two coalesced gather+sum passes.
+    flat_p = (t_int.view(T, 1) * N + p_idx.view(1, Bn)).reshape(-1)
+    flat_q = (t_int.view(T, 1) * N + q_idx.view(1, Bn)).reshape(-1)
+    both = torch.cat([flat_p, flat_q])              # (2·T·B,)
+    order = torch.argsort(both, stable=True)
+    Cnt = T * Bn * 2 // (Ts * N)                    # = Tdelta·(N−1)
+    gidx = order.reshape(Ts * N, Cnt)               # (Ts·N, Cnt)
     for _ in range(n_sweeps):
         # A^k for p-side rows: A = C_pq (J^k_q)^H ; for q-side rows:
         # A' = C_pq^H (J^k_p)^H  (from V_pq^H = Σ J^k_q C^H J_p^H)
@@ -110,19 +121,19 @@ def _solve_sweeps(data22, C22, J, rho_t, prox_target, p_idx, q_idx, N,
         rhs_q = abH_k2(V.mH, Wq)
         nm_p = abH_k2(Wp, Wp)                               # (F,T,B,2K,2K)
         nm_q = abH_k2(Wq, Wq)
-        # scatter-add into (F,Ts,N,…) by interval and station
-        rhs = V.new_zeros(F, Ts, N, 2, 2 * K)
-        nm = V.new_zeros(F, Ts, N, 2 * K, 2 * K)
-        flat_p = (t_int.view(T, 1) * N + p_idx.view(1, Bn)).reshape(-1)
-        flat_q = (t_int.view(T, 1) * N + q_idx.view(1, Bn)).reshape(-1)
-        rhs.view(F, Ts * N, 2, 2 * K).index_add_(
-            1, flat_p, rhs_p.reshape(F, T * Bn, 2, 2 * K))
-        rhs.view(F, Ts * N, 2, 2 * K).index_add_(
-            1, flat_q, rhs_q.reshape(F, T * Bn, 2, 2 * K))
-        nm.view(F, Ts * N, 2 * K, 2 * K).index_add_(
-            1, flat_p, nm_p.reshape(F, T * Bn, 2 * K, 2 * K))
-        nm.view(F, Ts * N, 2 * K, 2 * K).index_add_(
-            1, flat_q, nm_q.reshape(F, T * Bn, 2 * K, 2 * K))
+        # reduce into (F,Ts,N,…) by interval and station via the
+        # precomputed gather plan (see above)
+        rhs_cat = torch.cat([rhs_p.reshape(F, T * Bn, 2 * 2 * K),
+                             rhs_q.reshape(F, T * Bn, 2 * 2 * K)], dim=1)
+        nm_cat = torch.cat(
+            [nm_p.reshape(F, T * Bn, 2 * K * 2 * K),
+             nm_q.reshape(F, T * Bn, 2 * K * 2 * K)], dim=1)
+        rhs = rhs_cat[:, gidx.reshape(-1)] \
+            .reshape(F, Ts * N, Cnt, 2 * 2 * K).sum(dim=2) \
+            .reshape(F, Ts, N, 2, 2 * K)
+        nm = nm_cat[:, gidx.reshape(-1)] \
+            .reshape(F, Ts * N, Cnt, 2 * K * 2 * K).sum(dim=2) \
+            .reshape(F, Ts, N, 2 * K, 2 * K)
         # ADMM prox: + diag(ρ_k I2) and + ρ_k F^k on the rhs
         if prox_target is not None:
             rho_blocks = torch.kron(
