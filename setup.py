@@ -24,6 +24,7 @@ sources = [str(CSRC / f) for f in [
     "gemm_f32.hip",
     "elementwise.hip",
     "enet_solver.hip",
+    "als_sweep.hip",
 ]]
 
 setup(

@@ -39,6 +39,10 @@ extern "C" __global__ void adam_bump_kernel(float*);
 extern "C" __global__ void enet_lbfgs_solve_kernel(
     const float*, const float*, const float*, float*, float*, float*, int*,
     int, int, int, int, int, int);
+struct c32b { float x, y; };
+extern "C" __global__ void als_sweep_kernel(
+    const c32b*, const c32b*, const c32b*, const int*, const int*,
+    const int*, c32b*, c32b*, int, int, int, int, int, int);
 extern "C" __global__ void enet_influence_kernel(
     const float*, const float*, const float*, const float*, const float*,
     const int*, const float*, float*, float*, int, int, int);
@@ -179,6 +183,38 @@ mlp_chain_fwd(const at::Tensor& x,
   hipLaunchKernelGGL(mlp_chain_fwd_kernel, dim3((B + 15) / 16), dim3(1024),
                      lds_bytes, stream(), x.data_ptr<float>(), args);
   return {ys, zhats, rstds};
+}
+
+// Fused ALS-sweep contributions for the calibration solver: one launch
+// instead of ~100 elementwise kernels per sweep (radio/solver.py).
+std::tuple<at::Tensor, at::Tensor> als_sweep(
+    const at::Tensor& C22, const at::Tensor& V22, const at::Tensor& J,
+    const at::Tensor& p_idx, const at::Tensor& q_idx,
+    const at::Tensor& t_int) {
+  TORCH_CHECK(C22.is_cuda() && C22.scalar_type() == at::kComplexFloat
+              && C22.is_contiguous(), "C22 must be contiguous cfloat GPU");
+  TORCH_CHECK(V22.is_contiguous() && J.is_contiguous(), "contiguous");
+  TORCH_CHECK(p_idx.scalar_type() == at::kInt, "p_idx int32");
+  const int F = C22.size(0), K = C22.size(1), T = C22.size(2),
+            B = C22.size(3);
+  const int Ts = J.size(1), N = J.size(3);
+  TORCH_CHECK(K <= 8, "als_sweep supports K <= 8");
+  auto rhs_cat = at::empty({F, 2L * T * B, 2L * 2 * K},
+                           C22.options());
+  auto nm_cat = at::empty({F, 2L * T * B, 4L * K * K}, C22.options());
+  const long total = (long)F * T * B;
+  hipLaunchKernelGGL(als_sweep_kernel,
+                     dim3((unsigned)((total + 255) / 256)), dim3(256), 0,
+                     stream(),
+                     reinterpret_cast<const c32b*>(C22.data_ptr()),
+                     reinterpret_cast<const c32b*>(V22.data_ptr()),
+                     reinterpret_cast<const c32b*>(J.data_ptr()),
+                     p_idx.data_ptr<int>(), q_idx.data_ptr<int>(),
+                     t_int.data_ptr<int>(),
+                     reinterpret_cast<c32b*>(rhs_cat.data_ptr()),
+                     reinterpret_cast<c32b*>(nm_cat.data_ptr()),
+                     F, K, T, B, N, Ts);
+  return {rhs_cat, nm_cat};
 }
 
 // Direct-accumulate variant of fused_linear_bwd_dz: dgamma/dbeta are
@@ -358,6 +394,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_gemm_tn_bias_into", &mfma_gemm_tn_bias_into);
   m.def("fused_linear_bwd_dz_into", &fused_linear_bwd_dz_into);
   m.def("mlp_chain_fwd", &mlp_chain_fwd);
+  m.def("als_sweep", &als_sweep);
   m.def("tanh_gauss_fwd", &tanh_gauss_fwd);
   m.def("tanh_gauss_bwd", &tanh_gauss_bwd);
   m.def("fused_adam", &fused_adam);

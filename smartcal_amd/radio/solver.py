@@ -100,34 +100,49 @@ def _solve_sweeps(data22, C22, J, rho_t, prox_target, p_idx, q_idx, N,
     order = torch.argsort(both, stable=True)
     Cnt = T * Bn * 2 // (Ts * N)                    # = Tdelta·(N−1)
     gidx = order.reshape(Ts * N, Cnt)               # (Ts·N, Cnt)
+    use_kernel = dev.type == "cuda" and K <= 8
+    if use_kernel:
+        from ..ops import ext
+        p32 = p_idx.to(torch.int32).contiguous()
+        q32 = q_idx.to(torch.int32).contiguous()
+        t32 = t_int.to(torch.int32).contiguous()
+        C22c = C22.contiguous()
     for _ in range(n_sweeps):
-        # A^k for p-side rows: A = C_pq (J^k_q)^H ; for q-side rows:
-        # A' = C_pq^H (J^k_p)^H  (from V_pq^H = Σ J^k_q C^H J_p^H)
-        Jt = J[:, t_int]                                   # (F,T,K,N,2,2)
-        Jq = Jt[:, :, :, q_idx]                            # (F,T,K,B,2,2)
-        Jp = Jt[:, :, :, p_idx]
-        Cp = C22.permute(0, 2, 3, 1, 4, 5)                 # (F,T,B,K,2,2)
-        # elementwise small-complex products (see radio.small_complex):
-        # batched 2×2 / k=2 shapes are pathological for rocBLAS
-        from .small_complex import mm2H, Hmm2, abH_k2
-        A_p = mm2H(Cp, Jq.permute(0, 1, 3, 2, 4, 5))       # (F,T,B,K,2,2)
-        A_q = Hmm2(Cp, Jp.permute(0, 1, 3, 2, 4, 5).conj().mT)
-        V = data22                                          # (F,T,B,2,2)
-        # accumulate per (station, interval): normal matrix (2K,2K), rhs (2,2K)
-        # W = A stacked over k → (2K,2) ; contribution V·W^H (2,2K), W·W^H
-        Wp = A_p.reshape(F, T, Bn, 2 * K, 2)
-        Wq = A_q.reshape(F, T, Bn, 2 * K, 2)
-        rhs_p = abH_k2(V, Wp)                               # (F,T,B,2,2K)
-        rhs_q = abH_k2(V.mH, Wq)
-        nm_p = abH_k2(Wp, Wp)                               # (F,T,B,2K,2K)
-        nm_q = abH_k2(Wq, Wq)
+        if use_kernel:
+            # fused HIP kernel: all per-sample Jones products +
+            # normal-equation contributions in ONE launch (the env step
+            # is dispatch-bound; see ops/csrc/als_sweep.hip)
+            rhs_cat, nm_cat = ext().als_sweep(
+                C22c, data22.contiguous(), J.contiguous(), p32, q32, t32)
+        else:
+            # A^k for p-side rows: A = C_pq (J^k_q)^H ; for q-side rows:
+            # A' = C_pq^H (J^k_p)^H  (from V_pq^H = Σ J^k_q C^H J_p^H)
+            Jt = J[:, t_int]                               # (F,T,K,N,2,2)
+            Jq = Jt[:, :, :, q_idx]                        # (F,T,K,B,2,2)
+            Jp = Jt[:, :, :, p_idx]
+            Cp = C22.permute(0, 2, 3, 1, 4, 5)             # (F,T,B,K,2,2)
+            # elementwise small-complex products (radio.small_complex):
+            # batched 2×2 / k=2 shapes are pathological for rocBLAS
+            from .small_complex import mm2H, Hmm2, abH_k2
+            A_p = mm2H(Cp, Jq.permute(0, 1, 3, 2, 4, 5))   # (F,T,B,K,2,2)
+            A_q = Hmm2(Cp, Jp.permute(0, 1, 3, 2, 4, 5).conj().mT)
+            V = data22                                      # (F,T,B,2,2)
+            # per (station, interval): normal matrix (2K,2K), rhs (2,2K)
+            # W = A stacked over k → (2K,2); contribs V·W^H and W·W^H
+            Wp = A_p.reshape(F, T, Bn, 2 * K, 2)
+            Wq = A_q.reshape(F, T, Bn, 2 * K, 2)
+            rhs_p = abH_k2(V, Wp)                           # (F,T,B,2,2K)
+            rhs_q = abH_k2(V.mH, Wq)
+            nm_p = abH_k2(Wp, Wp)                           # (F,T,B,2K,2K)
+            nm_q = abH_k2(Wq, Wq)
+            rhs_cat = torch.cat(
+                [rhs_p.reshape(F, T * Bn, 2 * 2 * K),
+                 rhs_q.reshape(F, T * Bn, 2 * 2 * K)], dim=1)
+            nm_cat = torch.cat(
+                [nm_p.reshape(F, T * Bn, 2 * K * 2 * K),
+                 nm_q.reshape(F, T * Bn, 2 * K * 2 * K)], dim=1)
         # reduce into (F,Ts,N,…) by interval and station via the
         # precomputed gather plan (see above)
-        rhs_cat = torch.cat([rhs_p.reshape(F, T * Bn, 2 * 2 * K),
-                             rhs_q.reshape(F, T * Bn, 2 * 2 * K)], dim=1)
-        nm_cat = torch.cat(
-            [nm_p.reshape(F, T * Bn, 2 * K * 2 * K),
-             nm_q.reshape(F, T * Bn, 2 * K * 2 * K)], dim=1)
         rhs = rhs_cat[:, gidx.reshape(-1)] \
             .reshape(F, Ts * N, Cnt, 2 * 2 * K).sum(dim=2) \
             .reshape(F, Ts, N, 2, 2 * K)
