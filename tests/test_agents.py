@@ -99,3 +99,38 @@ def test_polyak_flat_matches_reference_rule():
     agent.update_network_parameters()  # tau = 0.3
     expected = 0.3 * online + 0.7 * target0
     assert torch.allclose(agent.target_critic_fp.flat, expected, atol=1e-6)
+
+
+def test_seeded_determinism():
+    """Same seed → identical scores (golden-run regression guard)."""
+    import numpy as np
+    import torch
+    from smartcal_amd.envs.enet import ENetEnv
+    from smartcal_amd.rl.sac import Agent
+    from smartcal_amd.utils.device import seed_everything
+
+    def run():
+        seed_everything(7)
+        env = ENetEnv(6, 6, device=torch.device("cpu"))
+        agent = Agent(gamma=0.99, batch_size=4, n_actions=2, tau=0.005,
+                      max_mem_size=32, input_dims=[6 + 36], lr_a=1e-3,
+                      lr_c=1e-3, reward_scale=6, alpha=0.03,
+                      device=torch.device("cpu"))
+        scores = []
+        for _ in range(3):
+            obs = env.reset()
+            tot = 0.0
+            for _ in range(3):
+                a = agent.choose_action(obs)
+                obs2, r, done, info = env.step(a)
+                agent.store_transition(obs, a, r, obs2, done,
+                                       np.zeros(2, np.float32))
+                agent.learn()
+                tot += float(r)
+                obs = obs2
+            scores.append(tot)
+        return scores
+
+    s1 = run()
+    s2 = run()
+    np.testing.assert_allclose(s1, s2, rtol=0, atol=0)
