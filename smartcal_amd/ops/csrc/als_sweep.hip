@@ -38,8 +38,8 @@ extern "C" __global__ void als_sweep_kernel(
     const int* __restrict__ p_idx, // (B,)
     const int* __restrict__ q_idx, // (B,)
     const int* __restrict__ t_int, // (T,)
-    c32* __restrict__ rhs_cat,     // (F, 2*T*B, 2*2K)
-    c32* __restrict__ nm_cat,      // (F, 2*T*B, 2K*2K)
+    c32* __restrict__ rhs_cat,     // (F, 2*2K, 2*T*B) — entry-major so
+    c32* __restrict__ nm_cat,      // (F, 2K*2K, 2*T*B)  writes coalesce
     int F, int K, int T, int B, int N, int Ts) {
   const long s = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long TB = (long)T * B;
@@ -93,11 +93,12 @@ extern "C" __global__ void als_sweep_kernel(
       }
   }
 
-  // output rows: p side at row tb, q side at row TB + tb
-  const long rowp = (long)f * (2 * TB) + tb;
-  const long rowq = rowp + TB;
-  c32* rp = rhs_cat + rowp * (2 * K2);
-  c32* rq = rhs_cat + rowq * (2 * K2);
+  // output columns: p side at col tb, q side at col TB + tb; entry e of
+  // sample col lives at ((f*X + e) * 2TB + col) so adjacent lanes write
+  // adjacent addresses (coalesced)
+  const long TB2 = 2 * TB;
+  c32* rp = rhs_cat + (long)f * (2 * K2) * TB2 + tb;
+  c32* rq = rp + TB;
   // rhs_p[i][c=2k+j] = sum_t V[i][t] conj(W_p[c][t]),  W_p[2k+j][t]=Ap[k][j][t]
 #pragma unroll
   for (int k = 0; k < KMAX; ++k) {
@@ -108,18 +109,18 @@ extern "C" __global__ void als_sweep_kernel(
       for (int i = 0; i < 2; ++i) {
         c32 vp = cadd(cmulj(V[i][0], Ap[k][j][0]),
                       cmulj(V[i][1], Ap[k][j][1]));
-        rp[i * K2 + 2 * k + j] = vp;
+        rp[(long)(i * K2 + 2 * k + j) * TB2] = vp;
         // rhs_q uses V^H: (V^H)[i][t] = conj(V[t][i])
         // rhs_q[i][c] = sum_t conj(V[t][i]) * conj(Aq[c][t])
         c32 vq = cadd(cmulj((c32){V[0][i].x, -V[0][i].y}, Aq[k][j][0]),
                       cmulj((c32){V[1][i].x, -V[1][i].y}, Aq[k][j][1]));
-        rq[i * K2 + 2 * k + j] = vq;
+        rq[(long)(i * K2 + 2 * k + j) * TB2] = vq;
       }
     }
   }
   // nm[r][c] = sum_t W[r][t] conj(W[c][t]) — Hermitian, compute full
-  c32* np_ = nm_cat + rowp * (K2 * K2);
-  c32* nq_ = nm_cat + rowq * (K2 * K2);
+  c32* np_ = nm_cat + (long)f * (K2 * K2) * TB2 + tb;
+  c32* nq_ = np_ + TB;
 #pragma unroll
   for (int kr = 0; kr < KMAX; ++kr) {
     if (kr >= K) break;
@@ -132,10 +133,12 @@ extern "C" __global__ void als_sweep_kernel(
 #pragma unroll
         for (int jc = 0; jc < 2; ++jc) {
           const int c = 2 * kc + jc;
-          np_[r * K2 + c] = cadd(cmulj(Ap[kr][jr][0], Ap[kc][jc][0]),
-                                 cmulj(Ap[kr][jr][1], Ap[kc][jc][1]));
-          nq_[r * K2 + c] = cadd(cmulj(Aq[kr][jr][0], Aq[kc][jc][0]),
-                                 cmulj(Aq[kr][jr][1], Aq[kc][jc][1]));
+          np_[(long)(r * K2 + c) * TB2] =
+              cadd(cmulj(Ap[kr][jr][0], Ap[kc][jc][0]),
+                   cmulj(Ap[kr][jr][1], Ap[kc][jc][1]));
+          nq_[(long)(r * K2 + c) * TB2] =
+              cadd(cmulj(Aq[kr][jr][0], Aq[kc][jc][0]),
+                   cmulj(Aq[kr][jr][1], Aq[kc][jc][1]));
         }
       }
     }

@@ -199,9 +199,10 @@ std::tuple<at::Tensor, at::Tensor> als_sweep(
             B = C22.size(3);
   const int Ts = J.size(1), N = J.size(3);
   TORCH_CHECK(K <= 8, "als_sweep supports K <= 8");
-  auto rhs_cat = at::empty({F, 2L * T * B, 2L * 2 * K},
-                           C22.options());
-  auto nm_cat = at::empty({F, 2L * T * B, 4L * K * K}, C22.options());
+  // entry-major layout (F, X, 2TB): kernel writes coalesce across the
+  // sample axis and the gather+sum reduces along the last dim
+  auto rhs_cat = at::empty({F, 2L * 2 * K, 2L * T * B}, C22.options());
+  auto nm_cat = at::empty({F, 4L * K * K, 2L * T * B}, C22.options());
   const long total = (long)F * T * B;
   hipLaunchKernelGGL(als_sweep_kernel,
                      dim3((unsigned)((total + 255) / 256)), dim3(256), 0,
