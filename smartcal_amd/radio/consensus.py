@@ -82,3 +82,43 @@ def hessian_addition(Ne: int, N: int, freqs: np.ndarray, f0: float,
         Hadd = 0.5 * rho_spectral * np.kron(
             np.eye(2), FF @ (I2N + np.linalg.pinv(I2N - FF) @ FF))
     return Hadd.astype(np.float32)
+
+
+def hessian_addition_scalar(Ne: int, N: int, freqs: np.ndarray, f0: float,
+                            fidx: int, rho_spectral: float,
+                            rho_spatial: float, polytype: int = 1) -> float:
+    """Scalar c with Hadd = c·I₄N.
+
+    Every matrix in `hessian_addition` is a scalar multiple of I₂N
+    (F = I − ρ·kron(b_f Bi b_fᵀ, I) — the reference notes "F … is
+    diagonal scalar", `calibration_tools.py:583`), so the whole Schur
+    complement collapses to scalar arithmetic on the Ne×Ne Gram matrix —
+    O(Ne³) instead of a 2N×2N pinv per direction per step.
+    """
+    freqs = np.asarray(freqs, dtype=np.float64)
+    Nf = len(freqs)
+    alpha = rho_spatial
+    rho = rho_spectral
+    if polytype == 0:
+        Bfull = np.zeros((Nf, Ne))
+        Bfull[:, 0] = 1.0
+        ff = (freqs - f0) / f0
+        for cj in range(1, Ne):
+            Bfull[:, cj] = ff ** cj
+    else:
+        ff = (freqs - freqs.min()) / (freqs.max() - freqs.min())
+        Bfull = bpoly(ff, Ne - 1).astype(np.float64)
+    Bi = np.linalg.pinv(rho * (Bfull.T @ Bfull) + alpha * np.eye(Ne))
+    bf = Bfull[fidx]
+    s = float(bf @ Bi @ bf)
+    f1 = 1.0 - rho * s                  # F = f1·I
+    FF = f1 * f1
+    def _pinv_s(x):
+        return 1.0 / x if abs(x) > 1e-14 else 0.0
+    if alpha > 0.0:
+        p = float(bf @ Bi.T @ Bi @ bf)  # PᵀP = p·I
+        h11 = 0.5 * rho * FF + 0.5 * alpha * rho * rho * p
+        h12 = 0.5 * FF + 0.5 * alpha * rho * p
+        h22 = -0.5 / rho * (1.0 - FF) + 0.5 * alpha * p
+        return float(h11 - h12 * _pinv_s(h22) * h12)
+    return float(0.5 * rho * FF * (1.0 + _pinv_s(1.0 - FF) * FF))

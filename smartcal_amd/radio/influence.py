@@ -16,7 +16,7 @@ import numpy as np
 import torch
 
 from . import hessian as hs
-from .consensus import hessian_addition
+from .consensus import hessian_addition_scalar
 from .sim import VisData, to_R
 
 __all__ = ["influence_values", "influence_per_direction", "hadd_for"]
@@ -25,16 +25,19 @@ __all__ = ["influence_values", "influence_per_direction", "hadd_for"]
 def hadd_for(K: int, N: int, Ne: int, freqs: np.ndarray, f0: float,
              fidx: int, rho_spectral, rho_spatial, device,
              polytype: int = 1) -> torch.Tensor:
-    """(K,4N,4N) consensus Hessian additions (`analysis_torch.py:141-156`)."""
-    H = torch.zeros((K, 4 * N, 4 * N), dtype=torch.complex64, device=device)
-    for k in range(K):
-        H[k] = torch.as_tensor(
-            hessian_addition(Ne, N, freqs, f0, fidx,
-                             float(rho_spectral[k]),
-                             float(rho_spatial[k]) if rho_spatial is not None
-                             else 0.0, polytype=polytype),
-            device=device)
-    return H
+    """(K,4N,4N) consensus Hessian additions (`analysis_torch.py:141-156`).
+
+    Every per-direction addition is a scalar multiple of the identity
+    (see `consensus.hessian_addition_scalar`), so this is K scalar
+    evaluations + one batched diag-embed — no 2N×2N pinvs per step."""
+    scal = torch.tensor(
+        [hessian_addition_scalar(
+            Ne, N, freqs, f0, fidx, float(rho_spectral[k]),
+            float(rho_spatial[k]) if rho_spatial is not None else 0.0,
+            polytype=polytype) for k in range(K)],
+        dtype=torch.float32)
+    eye = torch.eye(4 * N, dtype=torch.complex64, device=device)
+    return scal.to(device).view(K, 1, 1).to(torch.complex64) * eye
 
 
 def influence_values(residual4: torch.Tensor, C: torch.Tensor,

@@ -256,3 +256,13 @@ t1, POINT, Target, 12:00:00.0, 45.00.00.0, 2.0, 0, 0, 0, 150e6, [0.1], , ,
     assert len(sky2) == 3
     cl2 = parse_cluster_text(cl_t)
     assert cl2[0].cid == 2
+
+
+def test_hessian_addition_scalar_matches_matrix():
+    freqs = np.linspace(115e6, 185e6, 8)
+    for rho, alpha in ((0.9, 0.0), (0.9, 0.5), (12.0, 3.0), (100.0, 5.0)):
+        H = consensus.hessian_addition(3, 4, freqs, 150e6, 2, rho, alpha)
+        c = consensus.hessian_addition_scalar(3, 4, freqs, 150e6, 2, rho,
+                                              alpha)
+        np.testing.assert_allclose(H, c * np.eye(16), rtol=2e-4,
+                                   atol=2e-4 * max(1, abs(c)))
