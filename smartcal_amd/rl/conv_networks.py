@@ -3,10 +3,18 @@
 Same architecture family as the reference CNN agents
 (`calibration/calib_sac.py:90-250`, `demixing_rl/demix_sac.py:372-518`):
 3 convs (1→16→32→32, k5 s2, BatchNorm) on the influence map, a 2-layer
-FC on metadata (+action for critics), concatenated into a head. Conv and
-norm layers run through PyTorch-ROCm (MIOpen kernels on gfx950); the
-surrounding agents keep parameters in flat pools so optimizer/polyak/
-all-reduce are single fused kernels.
+FC on metadata (+action for critics), concatenated into a head.
+
+Design note: the convolutions are plain dense convs with no fusion
+opportunity beyond conv+BN+act, and MIOpen ships tuned gfx950 direct/
+implicit-GEMM kernels for exactly these shapes — so they run through
+PyTorch-ROCm/MIOpen, the same "library kernels for plain library ops"
+rule that routes plain GEMMs to rocBLAS while hand-written HIP covers
+the fused/bespoke hot ops (MLP chains, solver, sampler, PER, radio
+math). The surrounding agents keep parameters in flat pools so the
+optimizer / polyak / DP all-reduce are single fused kernels, and the
+twin critic/target forwards overlap on side HIP streams
+(`utils.streams.StreamFork`).
 """
 
 from __future__ import annotations
