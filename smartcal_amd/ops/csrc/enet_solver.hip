@@ -456,9 +456,15 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
   if (lane < M) xs[lane] = xg[(long)env * M + lane];
   if (lane < N) ys_[lane] = yg[(long)env * N + lane];
 
-  // Q[m][c] = -2 * A[c][m]
-  for (int m = 0; m < M; ++m)
-    if (lane < N) Q[m * N + lane] = -2.f * A[lane * M + m];
+  // Q column cached per lane in REGISTERS: qc[m] = Q[m][lane] with
+  // Q = -2 A^T, i.e. qc[m] = -2 A[lane][m] (this lane\'s row of A).
+  // The two-loop recursion then runs on registers with one front-loaded
+  // LDS broadcast per history entry instead of an lgkmcnt wait per FMA.
+  __builtin_amdgcn_s_barrier();
+  float qc[32];
+#pragma unroll
+  for (int m = 0; m < 32; ++m)
+    qc[m] = (lane < N && m < M) ? -2.f * A[lane * M + m] : 0.f;
 
   float ys = 1.f, yy = 1.f;
   for (int i = 0; i < nh; ++i) {
@@ -469,40 +475,58 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
     ys = lds_dot(&Yv[(nh - 1) * M], &Sv[(nh - 1) * M], M);
     yy = lds_dot(&Yv[(nh - 1) * M], &Yv[(nh - 1) * M], M);
   }
+  __builtin_amdgcn_s_barrier();
 
-  // two-loop on the matrix of columns
-  for (int i = nh - 1; i >= 0; --i) {
-    if (lane < N) {
-      float s = 0.f;
-      for (int m = 0; m < M; ++m) s += Sv[i * M + m] * Q[m * N + lane];
-      alv[i * N + lane] = ro[i] * s;
+  // two-loop on the register-resident columns; al_r in registers
+  float al_r[HMAX];
+#pragma unroll
+  for (int i = HMAX - 1; i >= 0; --i) {
+    if (i >= nh) continue;
+    float sv[32], yv[32];
+#pragma unroll
+    for (int m = 0; m < 32; ++m) {
+      sv[m] = (m < M) ? Sv[i * M + m] : 0.f;
+      yv[m] = (m < M) ? Yv[i * M + m] : 0.f;
     }
-    for (int m = 0; m < M; ++m)
-      if (lane < N) Q[m * N + lane] -= Yv[i * M + m] * alv[i * N + lane];
+    float sacc = 0.f;
+#pragma unroll
+    for (int m = 0; m < 32; ++m) sacc += sv[m] * qc[m];
+    const float ai = ro[i] * sacc;
+    al_r[i] = ai;
+#pragma unroll
+    for (int m = 0; m < 32; ++m) qc[m] -= yv[m] * ai;
   }
   const float scale = (nh > 0) ? (ys / yy) : 1.f;
-  for (int m = 0; m < M; ++m)
-    if (lane < N) Q[m * N + lane] *= scale;
-  for (int i = 0; i < nh; ++i) {
-    float be = 0.f;
-    if (lane < N) {
-      float s = 0.f;
-      for (int m = 0; m < M; ++m) s += Yv[i * M + m] * Q[m * N + lane];
-      be = ro[i] * s;
+#pragma unroll
+  for (int m = 0; m < 32; ++m) qc[m] *= scale;
+#pragma unroll
+  for (int i = 0; i < HMAX; ++i) {
+    if (i >= nh) continue;
+    float sv[32], yv[32];
+#pragma unroll
+    for (int m = 0; m < 32; ++m) {
+      sv[m] = (m < M) ? Sv[i * M + m] : 0.f;
+      yv[m] = (m < M) ? Yv[i * M + m] : 0.f;
     }
-    for (int m = 0; m < M; ++m)
-      if (lane < N)
-        Q[m * N + lane] += Sv[i * M + m] * (alv[i * N + lane] - be);
+    float yacc = 0.f;
+#pragma unroll
+    for (int m = 0; m < 32; ++m) yacc += yv[m] * qc[m];
+    const float be = ro[i] * yacc;
+#pragma unroll
+    for (int m = 0; m < 32; ++m) qc[m] += sv[m] * (al_r[i] - be);
   }
 
-  // B = A @ R, symmetrized
+  // B = A @ R, symmetrized (R columns live in qc registers)
   for (int r = 0; r < N; ++r) {
-    if (lane < N) {
-      float s = 0.f;
-      for (int m = 0; m < M; ++m) s += A[r * M + m] * Q[m * N + lane];
-      B[r * N + lane] = s;
-    }
+    float ar[32];
+#pragma unroll
+    for (int m = 0; m < 32; ++m) ar[m] = (m < M) ? A[r * M + m] : 0.f;
+    float sacc = 0.f;
+#pragma unroll
+    for (int m = 0; m < 32; ++m) sacc += ar[m] * qc[m];
+    if (lane < N) B[r * N + lane] = sacc;
   }
+  __builtin_amdgcn_s_barrier();
   for (int r = 0; r < N; ++r)
     if (lane < N && lane > r) {
       float v = 0.5f * (B[r * N + lane] + B[lane * N + r]);
