@@ -25,6 +25,7 @@ sources = [str(CSRC / f) for f in [
     "elementwise.hip",
     "enet_solver.hip",
     "als_sweep.hip",
+    "conv2d.hip",
 ]]
 
 setup(

@@ -39,6 +39,14 @@ extern "C" __global__ void adam_bump_kernel(float*);
 extern "C" __global__ void enet_lbfgs_solve_kernel(
     const float*, const float*, const float*, float*, float*, float*, int*,
     int, int, int, int, int, int);
+extern "C" __global__ void conv2d_k5s2_fwd_kernel(
+    const float*, const float*, const float*, float*, int, int, int, int,
+    int, int, int);
+extern "C" __global__ void conv2d_k5s2_dx_kernel(
+    const float*, const float*, float*, int, int, int, int, int, int, int);
+extern "C" __global__ void conv2d_k5s2_dw_kernel(
+    const float*, const float*, float*, float*, int, int, int, int, int,
+    int, int);
 struct c32b { float x, y; };
 extern "C" __global__ void als_sweep_kernel(
     const c32b*, const c32b*, const c32b*, const int*, const int*,
@@ -183,6 +191,47 @@ mlp_chain_fwd(const at::Tensor& x,
   hipLaunchKernelGGL(mlp_chain_fwd_kernel, dim3((B + 15) / 16), dim3(1024),
                      lds_bytes, stream(), x.data_ptr<float>(), args);
   return {ys, zhats, rstds};
+}
+
+at::Tensor conv2d_k5s2_fwd(const at::Tensor& x, const at::Tensor& W,
+                           const c10::optional<at::Tensor>& bias) {
+  check_f32(x, "x");
+  check_f32(W, "W");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), Wd = x.size(3);
+  const int Cout = W.size(0);
+  TORCH_CHECK(W.size(1) == Cin && W.size(2) == 5 && W.size(3) == 5,
+              "conv2d_k5s2: weight must be (Cout, Cin, 5, 5)");
+  const int OH = (H - 5) / 2 + 1, OW = (Wd - 5) / 2 + 1;
+  auto y = at::empty({B, Cout, OH, OW}, x.options());
+  const long total = (long)B * Cout * OH * OW;
+  hipLaunchKernelGGL(conv2d_k5s2_fwd_kernel,
+                     dim3((unsigned)((total + 255) / 256)), dim3(256), 0,
+                     stream(), x.data_ptr<float>(), W.data_ptr<float>(),
+                     bias ? bias->data_ptr<float>() : nullptr,
+                     y.data_ptr<float>(), B, Cin, H, Wd, Cout, OH, OW);
+  return y;
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> conv2d_k5s2_bwd(
+    const at::Tensor& dy, const at::Tensor& x, const at::Tensor& W) {
+  check_f32(dy, "dy");
+  check_f32(x, "x");
+  check_f32(W, "W");
+  const int B = x.size(0), Cin = x.size(1), H = x.size(2), Wd = x.size(3);
+  const int Cout = W.size(0), OH = dy.size(2), OW = dy.size(3);
+  auto dx = at::empty_like(x);
+  auto dW = at::empty_like(W);
+  auto db = at::empty({Cout}, x.options());
+  const long totx = (long)B * Cin * H * Wd;
+  hipLaunchKernelGGL(conv2d_k5s2_dx_kernel,
+                     dim3((unsigned)((totx + 255) / 256)), dim3(256), 0,
+                     stream(), dy.data_ptr<float>(), W.data_ptr<float>(),
+                     dx.data_ptr<float>(), B, Cin, H, Wd, Cout, OH, OW);
+  hipLaunchKernelGGL(conv2d_k5s2_dw_kernel, dim3(Cout * Cin), dim3(256), 0,
+                     stream(), dy.data_ptr<float>(), x.data_ptr<float>(),
+                     dW.data_ptr<float>(), db.data_ptr<float>(), B, Cin, H,
+                     Wd, Cout, OH, OW);
+  return {dx, dW, db};
 }
 
 // Fused ALS-sweep contributions for the calibration solver: one launch
@@ -396,6 +445,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_linear_bwd_dz_into", &fused_linear_bwd_dz_into);
   m.def("mlp_chain_fwd", &mlp_chain_fwd);
   m.def("als_sweep", &als_sweep);
+  m.def("conv2d_k5s2_fwd", &conv2d_k5s2_fwd);
+  m.def("conv2d_k5s2_bwd", &conv2d_k5s2_bwd);
   m.def("tanh_gauss_fwd", &tanh_gauss_fwd);
   m.def("tanh_gauss_bwd", &tanh_gauss_bwd);
   m.def("fused_adam", &fused_adam);

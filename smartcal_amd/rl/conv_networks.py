@@ -5,16 +5,14 @@ Same architecture family as the reference CNN agents
 3 convs (1→16→32→32, k5 s2, BatchNorm) on the influence map, a 2-layer
 FC on metadata (+action for critics), concatenated into a head.
 
-Design note: the convolutions are plain dense convs with no fusion
-opportunity beyond conv+BN+act, and MIOpen ships tuned gfx950 direct/
-implicit-GEMM kernels for exactly these shapes — so they run through
-PyTorch-ROCm/MIOpen, the same "library kernels for plain library ops"
-rule that routes plain GEMMs to rocBLAS while hand-written HIP covers
-the fused/bespoke hot ops (MLP chains, solver, sampler, PER, radio
-math). The surrounding agents keep parameters in flat pools so the
-optimizer / polyak / DP all-reduce are single fused kernels, and the
-twin critic/target forwards overlap on side HIP streams
-(`utils.streams.StreamFork`).
+The convolutions are hand-written direct (im2col-free) CDNA4 kernels,
+forward and backward (`ops/csrc/conv2d.hip` via `ops.conv.FusedConv2d`)
+— these tiny-channel k5/s2 shapes favor direct per-output accumulation
+over MFMA implicit GEMM. BatchNorm stays a torch op (cheap reductions,
+exact running-stat semantics). The surrounding agents keep parameters
+in flat pools so the optimizer / polyak / DP all-reduce are single
+fused kernels, and the twin critic/target forwards overlap on side HIP
+streams (`utils.streams.StreamFork`).
 """
 
 from __future__ import annotations
@@ -25,6 +23,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 from torch.distributions import Normal
+
+from ..ops.conv import FusedConv2d
 
 EPS = 1e-6
 
@@ -49,11 +49,14 @@ class ConvEncoder(nn.Module):
 
     def __init__(self, h: int, w: int, act: str = "elu"):
         super().__init__()
-        self.conv1 = nn.Conv2d(1, 16, kernel_size=5, stride=2)
+        # hand-written direct-conv HIP kernels (ops/csrc/conv2d.hip) on
+        # GPU; F.conv2d on CPU. BatchNorm stays a torch op (cheap
+        # reductions; running-stat semantics preserved exactly).
+        self.conv1 = FusedConv2d(1, 16)
         self.bn1 = nn.BatchNorm2d(16)
-        self.conv2 = nn.Conv2d(16, 32, kernel_size=5, stride=2)
+        self.conv2 = FusedConv2d(16, 32)
         self.bn2 = nn.BatchNorm2d(32)
-        self.conv3 = nn.Conv2d(32, 32, kernel_size=5, stride=2)
+        self.conv3 = FusedConv2d(32, 32)
         self.bn3 = nn.BatchNorm2d(32)
         for c in (self.conv1, self.conv2, self.conv3):
             _init_layer(c)

@@ -298,3 +298,30 @@ def test_mlp_chain_matches_layers():
                                    mr.named_parameters()):
             torch.testing.assert_close(p.grad, pr.grad, rtol=2e-3,
                                        atol=2e-3)
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason='needs GPU')
+def test_conv2d_k5s2_matches_torch():
+    """Direct-conv HIP kernels ≡ F.conv2d (fwd + dx/dW/db)."""
+    from smartcal_amd.ops.conv import conv2d_k5s2
+    torch.manual_seed(0)
+    for (B, Cin, Cout, H) in ((4, 1, 16, 33), (3, 16, 32, 29),
+                              (2, 32, 32, 13)):
+        x = torch.randn(B, Cin, H, H, device="cuda", requires_grad=True)
+        W = torch.randn(Cout, Cin, 5, 5, device="cuda",
+                        requires_grad=True) * 0.1
+        W.retain_grad()
+        b = torch.randn(Cout, device="cuda", requires_grad=True)
+        b.retain_grad()
+        x2 = x.detach().clone().requires_grad_(True)
+        W2 = W.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        y = conv2d_k5s2(x, W, b)
+        y2 = F.conv2d(x2, W2, b2, stride=2)
+        torch.testing.assert_close(y, y2, rtol=2e-4, atol=2e-4)
+        g = torch.randn_like(y)
+        y.backward(g)
+        y2.backward(g)
+        torch.testing.assert_close(x.grad, x2.grad, rtol=2e-3, atol=2e-3)
+        torch.testing.assert_close(W.grad, W2.grad, rtol=2e-3, atol=2e-3)
+        torch.testing.assert_close(b.grad, b2.grad, rtol=2e-3, atol=2e-3)
