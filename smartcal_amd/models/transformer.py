@@ -20,6 +20,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.linear import FusedLinear
+
 __all__ = ["SupervisedBuffer", "TransformerEncoder",
            "scaled_dot_product", "MultiheadAttention", "EncoderBlock"]
 
@@ -77,7 +79,8 @@ def scaled_dot_product(q, k, v):
 
 class MultiheadAttention(nn.Module):
     """Fused-qkv MHA over the head axis of one sample
-    (`transformer_models.py:85-118`)."""
+    (`transformer_models.py:85-118`). Projections run on the
+    hand-written fused-linear HIP kernels (ops/csrc/fused_linear.hip)."""
 
     def __init__(self, input_dim: int, embed_dim: int, num_heads: int):
         super().__init__()
@@ -85,8 +88,10 @@ class MultiheadAttention(nn.Module):
         self.embed_dim = embed_dim
         self.num_heads = num_heads
         self.head_dim = embed_dim // num_heads
-        self.qkv_proj = nn.Linear(input_dim, 3 * embed_dim)
-        self.o_proj = nn.Linear(embed_dim, embed_dim)
+        self.qkv_proj = FusedLinear(input_dim, 3 * embed_dim, ln=False,
+                                    act="none")
+        self.o_proj = FusedLinear(embed_dim, embed_dim, ln=False,
+                                  act="none")
         nn.init.xavier_uniform_(self.qkv_proj.weight)
         self.qkv_proj.bias.data.fill_(0)
         nn.init.xavier_uniform_(self.o_proj.weight)
@@ -109,10 +114,12 @@ class EncoderBlock(nn.Module):
         super().__init__()
         self.self_attn = MultiheadAttention(input_dim, input_dim, num_heads)
         self.linear_net = nn.Sequential(
-            nn.Linear(input_dim, dim_feedforward),
+            FusedLinear(input_dim, dim_feedforward, ln=False, act="none",
+                        init_scale=1.0 / input_dim ** 0.5),
             nn.Dropout(dropout),
             nn.ReLU(inplace=True),
-            nn.Linear(dim_feedforward, input_dim),
+            FusedLinear(dim_feedforward, input_dim, ln=False, act="none",
+                        init_scale=1.0 / dim_feedforward ** 0.5),
         )
         self.norm1 = nn.LayerNorm(input_dim)
         self.norm2 = nn.LayerNorm(input_dim)
@@ -131,17 +138,23 @@ class TransformerEncoder(nn.Module):
     def __init__(self, num_layers, input_dim, model_dim, num_classes,
                  num_heads, dropout=0.0):
         super().__init__()
-        self.input_net = nn.Sequential(nn.Dropout(dropout),
-                                       nn.Linear(input_dim, model_dim))
+        # input_dim can exceed the fused kernel\'s 640-column cap only on
+        # the IN side (K is unbounded); FusedLinear handles any K
+        self.input_net = nn.Sequential(
+            nn.Dropout(dropout),
+            FusedLinear(input_dim, model_dim, ln=False, act="none",
+                        init_scale=1.0 / input_dim ** 0.5))
         self.layers = nn.ModuleList([
             EncoderBlock(model_dim, num_heads, model_dim, dropout)
             for _ in range(num_layers)])
         self.output_net = nn.Sequential(
-            nn.Linear(model_dim, model_dim),
+            FusedLinear(model_dim, model_dim, ln=False, act="none",
+                        init_scale=1.0 / model_dim ** 0.5),
             nn.LayerNorm(model_dim),
             nn.ReLU(inplace=True),
             nn.Dropout(dropout),
-            nn.Linear(model_dim, num_classes),
+            FusedLinear(model_dim, num_classes, ln=False, act="none",
+                        init_scale=1.0 / model_dim ** 0.5),
         )
 
     def forward(self, x):

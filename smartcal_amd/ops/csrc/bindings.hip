@@ -76,7 +76,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_fwd(
   check_f32(W, "W");
   const int B = x.size(0), K = x.size(1), N = W.size(0);
   TORCH_CHECK(W.size(1) == K, "W/K mismatch");
-  TORCH_CHECK(N <= 640, "fused_linear: N>640 unsupported (extend NT_MAX)");
+  TORCH_CHECK(N <= 1536, "fused_linear: N>1536 unsupported (extend NT_MAX)");
   TORCH_CHECK(!with_ln || gamma.has_value(), "LN requires gamma");
   auto y = at::empty({B, N}, x.options());
   auto zhat = with_ln ? at::empty({B, N}, x.options())

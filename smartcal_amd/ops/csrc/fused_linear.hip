@@ -22,7 +22,7 @@
 #include "common.h"
 
 #define NWAVES 8
-#define NT_MAX 5           // max 16-col tiles per wave => N <= 16*NWAVES*NT_MAX = 640
+#define NT_MAX 12          // max 16-col tiles per wave => N <= 16*NWAVES*NT_MAX = 1536
 #define BK 64              // K-tile staged in LDS
 #define WCH (16 * (BK / 4) / WAVE)  // float4 chunks per lane per W subtile (=4)
 
