@@ -117,7 +117,7 @@ class EncoderBlock(nn.Module):
             FusedLinear(input_dim, dim_feedforward, ln=False, act="none",
                         init_scale=1.0 / input_dim ** 0.5),
             nn.Dropout(dropout),
-            nn.ReLU(inplace=True),
+            nn.ReLU(),
             FusedLinear(dim_feedforward, input_dim, ln=False, act="none",
                         init_scale=1.0 / dim_feedforward ** 0.5),
         )
@@ -151,7 +151,7 @@ class TransformerEncoder(nn.Module):
             FusedLinear(model_dim, model_dim, ln=False, act="none",
                         init_scale=1.0 / model_dim ** 0.5),
             nn.LayerNorm(model_dim),
-            nn.ReLU(inplace=True),
+            nn.ReLU(),
             nn.Dropout(dropout),
             FusedLinear(model_dim, num_classes, ln=False, act="none",
                         init_scale=1.0 / model_dim ** 0.5),
