@@ -158,13 +158,27 @@ def make_demixing_sky(rng: np.random.Generator, f0: float = 150e6,
     """Demixing scenario: 5 A-team outliers + target field (target last,
     as in `generate_data.simulate_data` / `demixingenv.py`). Returns
     (sky, clusters (K=6, target last), separation, azimuth, elevation,
-    fluxes, ra0, dec0) with per-direction metadata in the reference's
-    units (sep/az/el in degrees? — the reference uses radians from
-    casacore 'rad' conversions; we keep radians)."""
-    # random target pointing at reasonable elevation
-    ra0 = rng.uniform(0, 2 * math.pi)
-    dec0 = rng.uniform(math.radians(20), math.radians(80))
-    lst = ra0 + rng.uniform(-0.3, 0.3)   # target near transit ± a bit
+    fluxes, ra0, dec0) with sep/az/el in radians (the units
+    `simulate_data` returns, `generate_data.py:891`).
+
+    Target selection mirrors `find_valid_target` strategy 1
+    (`generate_data.py:50-106`): the pointing is drawn within
+    0.5–30.5° of a random A-team source and accepted when the target is
+    at least 3° up at the (random) observing epoch — the regime where
+    demixing decisions actually matter."""
+    # find_valid_target-style rejection sampling
+    low_el = math.radians(3.0)
+    close_to = int(rng.integers(len(ATEAM)))
+    dist_max = math.radians(0.5 + 30 * rng.random())
+    while True:
+        ra0 = ATEAM[close_to][1] + rng.random() * dist_max
+        dec0 = ATEAM[close_to][2] + rng.random() * dist_max
+        dec0 = min(max(dec0, -math.pi / 2 + 0.01), math.pi / 2 - 0.01)
+        ra0 = ra0 % (2 * math.pi)
+        lst = rng.uniform(0, 2 * math.pi)   # random epoch
+        _, el0 = arr.azel_of(ra0, dec0, lst)
+        if el0 > low_el:
+            break
     names, ras, decs, sIs, sPs, clusters = [], [], [], [], [], []
     ateam = ATEAM if n_outliers is None else ATEAM[:n_outliers]
     K = len(ateam) + 1
