@@ -30,6 +30,7 @@ def train(use_hint: bool, seed: int = 5):
                   max_mem_size=4000, input_dims=(1, 128, 128),
                   meta_dim=3 * K + 2, lr_a=3e-4, lr_c=3e-4,
                   use_hint=use_hint, use_influence=False,
+                  hint_threshold=0.0, admm_rho=1.0,
                   device=torch.device("cuda"))
     scores = []
     t0 = time.time()
