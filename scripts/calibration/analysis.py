@@ -1,0 +1,68 @@
+"""Influence-map analysis CLI.
+
+The CLI form of `calibration/analysis_torch.py:186-215` operating on the
+framework's in-memory formats: a saved visibility npz (radio.io), a
+SAGECal solutions text file and sky/cluster/rho texts; writes the
+per-sample influence values (npz) and optionally a dirty influence
+image.
+"""
+
+import argparse
+import sys
+from pathlib import Path
+
+import numpy as np
+import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[2]))
+
+from smartcal_amd.radio import io as rio
+from smartcal_amd.radio import influence as rinf
+from smartcal_amd.radio import imaging as rimg
+from smartcal_amd.radio.coherency import predict_coherencies_uvw
+from smartcal_amd.radio.sky import (parse_sky_text, parse_cluster_text,
+                                    parse_rho_text)
+from smartcal_amd.radio.solutions import parse_solutions_text
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("skymodel")
+    ap.add_argument("clusterfile")
+    ap.add_argument("vis_npz", help="visibilities saved by radio.io")
+    ap.add_argument("rhofile")
+    ap.add_argument("solutionsfile")
+    ap.add_argument("--freq-index", default=0, type=int)
+    ap.add_argument("--poly-order", default=3, type=int)
+    ap.add_argument("--out", default="influence.npz")
+    ap.add_argument("--image", default=None, help="optional image npy")
+    ap.add_argument("--ninf", default=128, type=int)
+    args = ap.parse_args()
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    vis = rio.load_visdata(args.vis_npz, device=device)
+    sky = parse_sky_text(open(args.skymodel).read())
+    clusters = parse_cluster_text(open(args.clusterfile).read())
+    K = len(clusters)
+    rho_spec, rho_spat = parse_rho_text(open(args.rhofile).read(), K)
+    freq, J = parse_solutions_text(open(args.solutionsfile).read())
+    fi = args.freq_index
+    C = predict_coherencies_uvw(sky, clusters, vis.uvw, freq, vis.ra0,
+                                vis.dec0, smear_bw=180e3)
+    Jt = torch.as_tensor(J[:K], device=device)
+    Hadd = rinf.hadd_for(K, vis.N, args.poly_order, vis.freqs,
+                         float(np.mean(vis.freqs)), fi, rho_spec,
+                         rho_spat, device)
+    vals = rinf.influence_values(vis.data[fi], C, Jt, vis.N, vis.Tdelta,
+                                 Hadd)
+    np.savez_compressed(args.out, influence=vals.cpu().numpy())
+    print(f"wrote {args.out} ({vals.shape[0]} samples)")
+    if args.image:
+        sI = 0.5 * (vals[:, 0] + vals[:, 3])
+        img = rimg.dirty_image(vis.uvw, sI, freq, args.ninf)
+        np.save(args.image, img.cpu().numpy())
+        print(f"wrote {args.image}")
+
+
+if __name__ == "__main__":
+    main()

@@ -26,6 +26,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seed", default=0, type=int)
     ap.add_argument("--episodes", default=1000, type=int)
+    # reference alias (`demixing_rl/main_sac.py:21`): --iteration = episodes
+    ap.add_argument("--iteration", default=None, type=int,
+                    help="max episodes (alias of --episodes)")
     ap.add_argument("--steps", default=7, type=int)
     ap.add_argument("--use_hint", action="store_true", default=False)
     ap.add_argument("--load", action="store_true", default=False)
@@ -33,6 +36,8 @@ def main():
     ap.add_argument("--stations", default=62, type=int)
     ap.add_argument("--influence", action="store_true", default=False)
     args = ap.parse_args()
+    if args.iteration is not None:
+        args.episodes = args.iteration
     seed_everything(args.seed)
 
     K = 6

@@ -85,3 +85,40 @@ def test_distill_pipeline(tmp_path):
     r = _run("scripts/demixing_rl/train_tsk.py", "--iters", "5",
              "--batch", "2", cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_analysis_cli(tmp_path):
+    """End-to-end analysis CLI: vis npz + SAGECal text files → influence."""
+    import numpy as np
+    import torch
+    sys.path.insert(0, str(ROOT))
+    from smartcal_amd.radio import array as arr, sim, solver
+    from smartcal_amd.radio import io as rio
+    from smartcal_amd.radio.sky import (write_sky_text, write_cluster_text,
+                                        write_rho_text)
+    from smartcal_amd.radio.solutions import (J_to_solutions,
+                                              format_solutions_text)
+    rng = np.random.default_rng(2)
+    layout = arr.lofar_like_layout(N=6, rng=rng)
+    sky, cs, *_, ra0, dec0 = sim.make_demixing_sky(rng, n_outliers=1)
+    freqs = np.array([140e6, 150e6])
+    vis = sim.simulate_observation(layout, sky, cs, freqs, ra0, dec0,
+                                   Ts=1, Tdelta=4, snr=10, rng=rng,
+                                   torch_seed=0)
+    sol = solver.calibrate(vis, sky, cs, np.ones(2, np.float32),
+                           admm_iter=2, poly_order=2)
+    rio.save_visdata(vis, str(tmp_path / "vis.npz"))
+    (tmp_path / "sky.txt").write_text(write_sky_text(sky))
+    (tmp_path / "cluster.txt").write_text(write_cluster_text(cs))
+    (tmp_path / "rho.txt").write_text(write_rho_text([1.0, 1.0],
+                                                     [0.0, 0.0]))
+    a = J_to_solutions(sol.J_ref_layout(0).cpu().numpy(), vis.N)
+    (tmp_path / "sols.txt").write_text(
+        format_solutions_text(float(freqs[0]), vis.N, a))
+    r = _run("scripts/calibration/analysis.py", "sky.txt", "cluster.txt",
+             "vis.npz", "rho.txt", "sols.txt", "--image", "inf.npy",
+             "--ninf", "16", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (tmp_path / "influence.npz").exists()
+    img = np.load(tmp_path / "inf.npy")
+    assert img.shape == (16, 16) and np.isfinite(img).all()
