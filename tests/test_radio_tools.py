@@ -266,3 +266,26 @@ def test_hessian_addition_scalar_matches_matrix():
                                               alpha)
         np.testing.assert_allclose(H, c * np.eye(16), rtol=2e-4,
                                    atol=2e-4 * max(1, abs(c)))
+
+
+def test_parser_roundtrips_property():
+    """Property-style fuzz of the text round trips (hypothesis-lite)."""
+    rng = np.random.default_rng(42)
+    for trial in range(10):
+        ns = int(rng.integers(1, 12))
+        sm = _mk_sky(rng, ns=ns, gaussian_frac=rng.random())
+        sm2 = sky.parse_sky_text(sky.write_sky_text(sm))
+        np.testing.assert_allclose(sm2.ra, sm.ra, atol=1e-6)
+        np.testing.assert_allclose(sm2.sI, sm.sI, rtol=1e-4)
+        K = int(rng.integers(1, 5))
+        rs = rng.random(K) * 100
+        ra_ = rng.random(K)
+        rs2, ra2 = sky.parse_rho_text(sky.write_rho_text(rs, ra_), K)
+        np.testing.assert_allclose(rs2, rs, rtol=1e-5)
+        np.testing.assert_allclose(ra2, ra_, rtol=1e-4, atol=1e-5)
+        N, Nto, Kj = (int(rng.integers(2, 6)), int(rng.integers(1, 4)),
+                      int(rng.integers(1, 4)))
+        a = rng.standard_normal((8 * N * Nto, Kj)).astype(np.float32)
+        J = solutions.solutions_to_J(a, N, Nto)
+        np.testing.assert_allclose(solutions.J_to_solutions(J, N), a,
+                                   atol=1e-6)
