@@ -38,8 +38,8 @@ extern "C" __global__ void als_sweep_kernel(
     const int* __restrict__ p_idx, // (B,)
     const int* __restrict__ q_idx, // (B,)
     const int* __restrict__ t_int, // (T,)
-    float* __restrict__ rhs_out,   // (F, Ts, N, 2, 2K) c32, pre-zeroed
-    float* __restrict__ nm_out,    // (F, Ts, N, 2K, 2K) c32, pre-zeroed
+    c32* __restrict__ rhs_cat,     // (F, 2*2K, 2*T*B) — entry-major so
+    c32* __restrict__ nm_cat,      // (F, 2K*2K, 2*T*B)  writes coalesce
     int F, int K, int T, int B, int N, int Ts) {
   const long s = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long TB = (long)T * B;
@@ -93,13 +93,12 @@ extern "C" __global__ void als_sweep_kernel(
       }
   }
 
-  // accumulate straight into the (interval, station) bins with float
-  // atomics: the per-sample contribution tensors (0.5 GB/sweep at LOFAR
-  // scale) and the gather+sum reduction disappear entirely
-  const long baseP = ((((long)f * Ts + ti) * N + pp));
-  const long baseQ = ((((long)f * Ts + ti) * N + qq));
-  float* rp = rhs_out + baseP * (2 * K2) * 2;
-  float* rq = rhs_out + baseQ * (2 * K2) * 2;
+  // output columns: p side at col tb, q side at col TB + tb; entry e of
+  // sample col lives at ((f*X + e) * 2TB + col) so adjacent lanes write
+  // adjacent addresses (coalesced)
+  const long TB2 = 2 * TB;
+  c32* rp = rhs_cat + (long)f * (2 * K2) * TB2 + tb;
+  c32* rq = rp + TB;
   // rhs_p[i][c=2k+j] = sum_t V[i][t] conj(W_p[c][t]),  W_p[2k+j][t]=Ap[k][j][t]
 #pragma unroll
   for (int k = 0; k < KMAX; ++k) {
@@ -110,20 +109,18 @@ extern "C" __global__ void als_sweep_kernel(
       for (int i = 0; i < 2; ++i) {
         c32 vp = cadd(cmulj(V[i][0], Ap[k][j][0]),
                       cmulj(V[i][1], Ap[k][j][1]));
-        atomicAdd(rp + (i * K2 + 2 * k + j) * 2 + 0, vp.x);
-        atomicAdd(rp + (i * K2 + 2 * k + j) * 2 + 1, vp.y);
+        rp[(long)(i * K2 + 2 * k + j) * TB2] = vp;
         // rhs_q uses V^H: (V^H)[i][t] = conj(V[t][i])
         // rhs_q[i][c] = sum_t conj(V[t][i]) * conj(Aq[c][t])
         c32 vq = cadd(cmulj((c32){V[0][i].x, -V[0][i].y}, Aq[k][j][0]),
                       cmulj((c32){V[1][i].x, -V[1][i].y}, Aq[k][j][1]));
-        atomicAdd(rq + (i * K2 + 2 * k + j) * 2 + 0, vq.x);
-        atomicAdd(rq + (i * K2 + 2 * k + j) * 2 + 1, vq.y);
+        rq[(long)(i * K2 + 2 * k + j) * TB2] = vq;
       }
     }
   }
   // nm[r][c] = sum_t W[r][t] conj(W[c][t]) — Hermitian, compute full
-  float* np_ = nm_out + baseP * (K2 * K2) * 2;
-  float* nq_ = nm_out + baseQ * (K2 * K2) * 2;
+  c32* np_ = nm_cat + (long)f * (K2 * K2) * TB2 + tb;
+  c32* nq_ = np_ + TB;
 #pragma unroll
   for (int kr = 0; kr < KMAX; ++kr) {
     if (kr >= K) break;
@@ -136,14 +133,12 @@ extern "C" __global__ void als_sweep_kernel(
 #pragma unroll
         for (int jc = 0; jc < 2; ++jc) {
           const int c = 2 * kc + jc;
-          const c32 npv = cadd(cmulj(Ap[kr][jr][0], Ap[kc][jc][0]),
-                               cmulj(Ap[kr][jr][1], Ap[kc][jc][1]));
-          atomicAdd(np_ + (r * K2 + c) * 2 + 0, npv.x);
-          atomicAdd(np_ + (r * K2 + c) * 2 + 1, npv.y);
-          const c32 nqv = cadd(cmulj(Aq[kr][jr][0], Aq[kc][jc][0]),
-                               cmulj(Aq[kr][jr][1], Aq[kc][jc][1]));
-          atomicAdd(nq_ + (r * K2 + c) * 2 + 0, nqv.x);
-          atomicAdd(nq_ + (r * K2 + c) * 2 + 1, nqv.y);
+          np_[(long)(r * K2 + c) * TB2] =
+              cadd(cmulj(Ap[kr][jr][0], Ap[kc][jc][0]),
+                   cmulj(Ap[kr][jr][1], Ap[kc][jc][1]));
+          nq_[(long)(r * K2 + c) * TB2] =
+              cadd(cmulj(Aq[kr][jr][0], Aq[kc][jc][0]),
+                   cmulj(Aq[kr][jr][1], Aq[kc][jc][1]));
         }
       }
     }

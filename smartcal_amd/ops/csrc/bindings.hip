@@ -50,7 +50,7 @@ extern "C" __global__ void conv2d_k5s2_dw_kernel(
 struct c32b { float x, y; };
 extern "C" __global__ void als_sweep_kernel(
     const c32b*, const c32b*, const c32b*, const int*, const int*,
-    const int*, float*, float*, int, int, int, int, int, int);
+    const int*, c32b*, c32b*, int, int, int, int, int, int);
 extern "C" __global__ void enet_influence_kernel(
     const float*, const float*, const float*, const float*, const float*,
     const int*, const float*, float*, float*, int, int, int);
@@ -248,11 +248,10 @@ std::tuple<at::Tensor, at::Tensor> als_sweep(
             B = C22.size(3);
   const int Ts = J.size(1), N = J.size(3);
   TORCH_CHECK(K <= 8, "als_sweep supports K <= 8");
-  // reduced outputs: the kernel atomically accumulates straight into
-  // the (freq, interval, station) normal-equation bins
-  auto rhs = at::zeros({F, (long)Ts, (long)N, 2L, 2L * K}, C22.options());
-  auto nm = at::zeros({F, (long)Ts, (long)N, 2L * K, 2L * K},
-                      C22.options());
+  // entry-major layout (F, X, 2TB): kernel writes coalesce across the
+  // sample axis and the gather+sum reduces along the last dim
+  auto rhs_cat = at::empty({F, 2L * 2 * K, 2L * T * B}, C22.options());
+  auto nm_cat = at::empty({F, 4L * K * K, 2L * T * B}, C22.options());
   const long total = (long)F * T * B;
   hipLaunchKernelGGL(als_sweep_kernel,
                      dim3((unsigned)((total + 255) / 256)), dim3(256), 0,
@@ -262,10 +261,10 @@ std::tuple<at::Tensor, at::Tensor> als_sweep(
                      reinterpret_cast<const c32b*>(J.data_ptr()),
                      p_idx.data_ptr<int>(), q_idx.data_ptr<int>(),
                      t_int.data_ptr<int>(),
-                     reinterpret_cast<float*>(rhs.data_ptr()),
-                     reinterpret_cast<float*>(nm.data_ptr()),
+                     reinterpret_cast<c32b*>(rhs_cat.data_ptr()),
+                     reinterpret_cast<c32b*>(nm_cat.data_ptr()),
                      F, K, T, B, N, Ts);
-  return {rhs, nm};
+  return {rhs_cat, nm_cat};
 }
 
 // Direct-accumulate variant of fused_linear_bwd_dz: dgamma/dbeta are

@@ -110,11 +110,10 @@ def _solve_sweeps(data22, C22, J, rho_t, prox_target, p_idx, q_idx, N,
     for _ in range(n_sweeps):
         if use_kernel:
             # fused HIP kernel: all per-sample Jones products +
-            # normal-equation contributions accumulated atomically into
-            # the (freq, interval, station) bins in ONE launch
-            # (ops/csrc/als_sweep.hip) — no per-sample contribution
-            # tensors, no gather reduction
-            rhs, nm = ext().als_sweep(
+            # normal-equation contributions in ONE launch (the env step
+            # is dispatch-bound; see ops/csrc/als_sweep.hip). Output is
+            # entry-major (F, X, 2TB) for coalesced writes.
+            rhs_cat, nm_cat = ext().als_sweep(
                 C22c, data22.contiguous(), J.contiguous(), p32, q32, t32)
         else:
             # A^k for p-side rows: A = C_pq (J^k_q)^H ; for q-side rows:
@@ -145,14 +144,14 @@ def _solve_sweeps(data22, C22, J, rho_t, prox_target, p_idx, q_idx, N,
                 [nm_p.reshape(F, T * Bn, 2 * K * 2 * K),
                  nm_q.reshape(F, T * Bn, 2 * K * 2 * K)], dim=1) \
                 .permute(0, 2, 1)
-            # reduce into (F,Ts,N,…) by interval and station via the
-            # precomputed gather plan (see above); cat is (F, X, 2TB)
-            rhs = rhs_cat[:, :, gidx.reshape(-1)] \
-                .reshape(F, 2 * 2 * K, Ts * N, Cnt).sum(dim=3) \
-                .permute(0, 2, 1).reshape(F, Ts, N, 2, 2 * K)
-            nm = nm_cat[:, :, gidx.reshape(-1)] \
-                .reshape(F, 2 * K * 2 * K, Ts * N, Cnt).sum(dim=3) \
-                .permute(0, 2, 1).reshape(F, Ts, N, 2 * K, 2 * K)
+        # reduce into (F,Ts,N,…) by interval and station via the
+        # precomputed gather plan (see above); cat layout is (F, X, 2TB)
+        rhs = rhs_cat[:, :, gidx.reshape(-1)] \
+            .reshape(F, 2 * 2 * K, Ts * N, Cnt).sum(dim=3) \
+            .permute(0, 2, 1).reshape(F, Ts, N, 2, 2 * K)
+        nm = nm_cat[:, :, gidx.reshape(-1)] \
+            .reshape(F, 2 * K * 2 * K, Ts * N, Cnt).sum(dim=3) \
+            .permute(0, 2, 1).reshape(F, Ts, N, 2 * K, 2 * K)
         # ADMM prox: + diag(ρ_k I2) and + ρ_k F^k on the rhs
         if prox_target is not None:
             rho_blocks = torch.kron(
