@@ -29,6 +29,9 @@ def main():
     ap.add_argument("--learner-addr", type=str, default="localhost")
     ap.add_argument("--learner-port", type=int, default=6985)
     ap.add_argument("--episodes", type=int, default=1000)
+    ap.add_argument("--backend", default=None,
+                    help="nccl|gloo (default: nccl when each rank can "
+                         "own a GPU, else gloo)")
     args = ap.parse_args()
 
     def agent_factory():
@@ -40,12 +43,17 @@ def main():
     def env_factory():
         return ENetEnv(M, N, provide_hint=True)
 
+    import torch as _t
+    backend = args.backend
+    if backend is None:
+        backend = "nccl" if (_t.cuda.is_available()
+                             and _t.cuda.device_count() >= args.world_size)            else "gloo"
     run_process(args.rank, args.world_size, agent_factory, env_factory,
                 obs_dim=N + N * M, n_actions=2, episodes=args.episodes,
                 epochs=10, steps=10, use_hint=True,
                 learner_addr=args.learner_addr,
                 learner_port=args.learner_port, max_transitions=100,
-                save_every=10)
+                save_every=10, backend=backend)
 
 
 if __name__ == "__main__":

@@ -67,3 +67,5 @@ def use_hip(t: torch.Tensor) -> bool:
 from . import linear, sampling, polyak, enet, per  # noqa: E402,F401
 from .linear import fused_linear  # noqa: E402,F401
 from .sampling import tanh_gauss_sample  # noqa: E402,F401
+
+from . import conv  # noqa: E402,F401
