@@ -71,8 +71,9 @@ def main():
     import torch as _t
     backend = args.backend
     if backend is None:
-        backend = "nccl" if (_t.cuda.is_available()
-                             and _t.cuda.device_count() >= args.world_size)            else "gloo"
+        gpu_ok = (_t.cuda.is_available()
+                  and _t.cuda.device_count() >= args.world_size)
+        backend = "nccl" if gpu_ok else "gloo"
     run_process(args.rank, args.world_size, agent_factory, env_factory,
                 obs_dim=OBS_DIM, n_actions=K, episodes=args.episodes,
                 epochs=10, steps=7, use_hint=True,

@@ -46,8 +46,9 @@ def main():
     import torch as _t
     backend = args.backend
     if backend is None:
-        backend = "nccl" if (_t.cuda.is_available()
-                             and _t.cuda.device_count() >= args.world_size)            else "gloo"
+        gpu_ok = (_t.cuda.is_available()
+                  and _t.cuda.device_count() >= args.world_size)
+        backend = "nccl" if gpu_ok else "gloo"
     run_process(args.rank, args.world_size, agent_factory, env_factory,
                 obs_dim=N + N * M, n_actions=2, episodes=args.episodes,
                 epochs=10, steps=10, use_hint=True,
