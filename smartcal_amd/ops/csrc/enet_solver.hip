@@ -114,7 +114,10 @@ __device__ static float ls_eval(const LdsLayout& L, const RegA& Ar,
                                 float rho1, float rho2, float t,
                                 float* gtd_out) {
   const int lane = threadIdx.x;
-  if (lane < M) L.x[lane] = L.x0[lane] + t * L.d[lane];
+  if (lane < M) {
+    float xd = t * L.d[lane];
+    L.x[lane] = L.x0[lane] + (isfinite(xd) ? xd : 0.f);
+  }
   float f = eval_loss_grad(L, Ar, N, M, rho1, rho2);
   *gtd_out = lds_dot(L.g, L.d, M);
   return f;
@@ -122,15 +125,21 @@ __device__ static float ls_eval(const LdsLayout& L, const RegA& Ar,
 
 __device__ static float cubic_interp(float x1, float f1, float g1, float x2,
                                      float f2, float g2, float lo, float hi) {
+  // NaN/degenerate guards mirroring the reference line search\'s checks
+  // (`lbfgsnew.py:556,624,673`): fall back to the bisection midpoint
+  if (fabsf(x1 - x2) < 1e-20f || !isfinite(f1) || !isfinite(f2) ||
+      !isfinite(g1) || !isfinite(g2))
+    return 0.5f * (lo + hi);
   float d1 = g1 + g2 - 3.f * (f1 - f2) / (x1 - x2);
   float d2sq = d1 * d1 - g1 * g2;
-  if (d2sq >= 0.f) {
+  if (isfinite(d2sq) && d2sq >= 0.f) {
     float d2 = sqrtf(d2sq);
     float mp;
     if (x1 <= x2)
       mp = x2 - (x2 - x1) * ((g2 + d2 - d1) / (g2 - g1 + 2.f * d2));
     else
       mp = x1 - (x1 - x2) * ((g1 + d2 - d1) / (g1 - g2 + 2.f * d2));
+    if (!isfinite(mp)) return 0.5f * (lo + hi);
     return fminf(fmaxf(mp, lo), hi);
   }
   return 0.5f * (lo + hi);
@@ -163,6 +172,13 @@ __device__ static float strong_wolfe(const LdsLayout& L, const RegA& Ar,
   bool bracketed = false;
   while (ls_iter < max_ls) {
     f_new = ls_eval(L, Ar, N, M, rho1, rho2, t, &gtd_new);
+    if (!isfinite(f_new) || !isfinite(gtd_new) || !isfinite(t)) {
+      // blown-up step: restore the start point (reference NaN guard)
+      t = 0.f;
+      f_new = ls_eval(L, Ar, N, M, rho1, rho2, 0.f, &gtd_new);
+      *t_io = 0.f;
+      return f_new;
+    }
     if (f_new > (f0 + c1 * t * gtd0) || (ls_iter > 0 && f_new >= f_prev)) {
       br_t[0] = t_prev; br_f[0] = f_prev; br_gtd[0] = gtd_prev;
       // bg0 already holds g(t_prev)
@@ -223,6 +239,12 @@ __device__ static float strong_wolfe(const LdsLayout& L, const RegA& Ar,
       insuf = false;
     }
     f_new = ls_eval(L, Ar, N, M, rho1, rho2, t, &gtd_new);
+    if (!isfinite(f_new) || !isfinite(gtd_new)) {
+      t = 0.f;
+      f_new = ls_eval(L, Ar, N, M, rho1, rho2, 0.f, &gtd_new);
+      *t_io = 0.f;
+      return f_new;
+    }
     if (f_new > (f0 + c1 * t * gtd0) || f_new >= br_f[low]) {
       br_t[high] = t; br_f[high] = f_new; br_gtd[high] = gtd_new;
       if (lane < M) br_g[high][lane] = L.g[lane];
@@ -247,7 +269,8 @@ __device__ static float strong_wolfe(const LdsLayout& L, const RegA& Ar,
     f_new = br_f[low];
     if (lane < M) {
       L.g[lane] = br_g[low][lane];
-      L.x[lane] = L.x0[lane] + t * L.d[lane];
+      float xd = t * L.d[lane];
+      L.x[lane] = L.x0[lane] + (isfinite(xd) ? xd : 0.f);
     }
   }
   // done==true: x and g already hold the accepted point
@@ -378,11 +401,20 @@ extern "C" __global__ __launch_bounds__(64) void enet_lbfgs_solve_kernel(
           if (lane < M) L.d[lane] += (L.al[i] - be) * L.S[i * M + lane];
         }
       }
+      {
+        float bad = wave_sum((lane < M && !isfinite(L.d[lane])) ? 1.f : 0.f);
+        if (bad > 0.f) {
+          if (lane < M) L.d[lane] = -L.g[lane];
+          nhist = 0;
+          H_diag = 1.f;
+        }
+      }
       if (lane < M) L.gprev[lane] = L.g[lane];
       prev_loss = loss;
 
       float gtd = lds_dot(L.g, L.d, M);
-      if (gtd > -TOL_CHANGE) { outer_done = true; break; }
+      // negated form so a NaN gtd also exits instead of entering the search
+      if (!(gtd <= -TOL_CHANGE)) { outer_done = true; break; }
 
       if (n_iter == 1) {
         float gsum = wave_sum((lane < M) ? fabsf(L.g[lane]) : 0.f);
@@ -393,6 +425,14 @@ extern "C" __global__ __launch_bounds__(64) void enet_lbfgs_solve_kernel(
 
       if (lane < M) L.x0[lane] = L.x[lane];
       loss = strong_wolfe(L, Ar, N, M, rho1, rho2, loss, gtd, &t);
+      if (!isfinite(loss)) {
+        // restore the pre-line-search point and stop (reference
+        // behaviour: NaN warning + early exit, lbfgsnew.py:695-714)
+        if (lane < M) L.x[lane] = L.x0[lane];
+        loss = eval_loss_grad(L, Ar, N, M, rho1, rho2);
+        outer_done = true;
+        break;
+      }
 
       if (lds_absmax(L.g, M) <= TOL_GRAD) { outer_done = true; break; }
       float step_max = lds_absmax(L.d, M) * fabsf(t);
