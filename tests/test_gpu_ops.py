@@ -325,3 +325,37 @@ def test_conv2d_k5s2_matches_torch():
         torch.testing.assert_close(x.grad, x2.grad, rtol=2e-3, atol=2e-3)
         torch.testing.assert_close(W.grad, W2.grad, rtol=2e-3, atol=2e-3)
         torch.testing.assert_close(b.grad, b2.grad, rtol=2e-3, atol=2e-3)
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason='needs GPU')
+def test_enet_solver_finite_small_rho():
+    """Regression: small-rho (ill-conditioned) instances must stay finite.
+
+    A seed-2 hint-path SAC episode drove rho to ~(0.07, 0.06) on an
+    instance where the in-kernel two-loop overflowed (1/ys) and the
+    unguarded line search emitted NaN x, poisoning the replay buffer.
+    The solver now carries the reference's NaN guards
+    (`lbfgsnew.py:556,624,673,695`); sweep the degenerate regime and
+    assert everything stays finite.
+    """
+    from smartcal_amd.envs.enet import ENetEnv
+    from smartcal_amd.utils.device import seed_everything
+    from smartcal_amd import ops
+
+    seed_everything(2)
+    env = ENetEnv(20, 20, provide_hint=True)
+    env.reset()
+    for rho1 in (1e-3, 0.0621, 0.0696, 0.5, 5.0):
+        for rho2 in (1e-3, 0.0621, 0.5):
+            x, EE, r = ops.enet.solve_and_influence(
+                env.A, env.y0, float(rho1), float(rho2), 0.0)
+            assert torch.isfinite(x).all(), (rho1, rho2)
+            assert torch.isfinite(EE).all(), (rho1, rho2)
+            assert np.isfinite(float(r)), (rho1, rho2)
+    # and through the env step API (device path) with adversarial actions
+    obs = env.reset()
+    for a in ([-1.0, -1.0], [0.3863, 0.2346], [1.0, 1.0], [-0.99, 0.99]):
+        act = torch.tensor(a, device="cuda")
+        out = env.step(act)
+        r = float(out[1])
+        assert np.isfinite(r), a
