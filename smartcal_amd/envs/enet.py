@@ -213,3 +213,42 @@ def obs_to_state(observation) -> torch.Tensor:
     if not torch.is_tensor(A):
         A = torch.as_tensor(A, dtype=torch.float32)
     return torch.cat((eig.reshape(-1), A.reshape(-1).to(eig.device)))
+
+class SKEnet:
+    """sklearn-compatible elastic-net estimator (reference
+    ``elasticnet/enetenv.py:249-295``): exposes fit/predict/score plus
+    get_params/set_params so it plugs into ``sklearn.model_selection
+    .GridSearchCV`` exactly like the reference's wrapper. The fit itself
+    runs the in-framework L-BFGS solver (CPU reference path, so it works
+    without a GPU) instead of scipy L-BFGS-B."""
+
+    def __init__(self, lambda1=0.001, lambda2=0.001):
+        self.lambda1 = lambda1  # L1 weight
+        self.lambda2 = lambda2  # L2 weight
+        self.x_ = None
+
+    def get_params(self, deep=True):
+        return {"lambda1": self.lambda1, "lambda2": self.lambda2}
+
+    def set_params(self, **params):
+        for k, v in params.items():
+            setattr(self, k, v)
+        return self
+
+    def fit(self, A, y):
+        A = torch.as_tensor(np.asarray(A), dtype=torch.float32)
+        y = torch.as_tensor(np.asarray(y), dtype=torch.float32).reshape(-1)
+        self.x_, _ = enet_ops.lbfgs_solve_reference(
+            A, y, rho1=float(self.lambda2), rho2=float(self.lambda1),
+            epochs=10, max_iter=10)
+        return self
+
+    def predict(self, A):
+        A = torch.as_tensor(np.asarray(A), dtype=torch.float32)
+        return (A @ self.x_).cpu().numpy()
+
+    def score(self, A, y):
+        """Negative MSE (higher is better), as sklearn scorers expect."""
+        r = self.predict(A) - np.asarray(y).reshape(-1)
+        return -float(np.mean(r * r))
+

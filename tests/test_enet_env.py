@@ -96,3 +96,20 @@ def test_env_hint_in_action_space():
         np.zeros(2, dtype=np.float32))
     assert hint.shape == (2,)
     assert (hint >= -1.01).all() and (hint <= 1.01).all()
+
+
+def test_skenet_gridsearchcv():
+    """SKEnet plugs into real sklearn GridSearchCV (`enetenv.py:249-295`)."""
+    import numpy as np
+    from sklearn.model_selection import GridSearchCV
+    from smartcal_amd.envs.enet import SKEnet
+    rng = np.random.default_rng(0)
+    A = rng.normal(size=(20, 20)).astype("f")
+    x = np.zeros(20, "f"); x[3] = 1.0
+    y = A @ x
+    gs = GridSearchCV(SKEnet(), {"lambda1": [0.001, 0.01],
+                                 "lambda2": [0.001, 0.01]}, cv=2)
+    gs.fit(A, y)
+    assert gs.best_score_ > -0.01
+    est = SKEnet(**gs.best_params_).fit(A, y)
+    assert np.mean((est.predict(A) - y) ** 2) < 1e-3

@@ -36,6 +36,7 @@ def _run(script, *args, cwd):
     "scripts/demixing_rl/evaluate_models.py",
     "scripts/elasticnet/distributed_per_sac.py",
     "scripts/demixing_rl/distributed_per_sac.py",
+    "scripts/calibration/inspect_replaybuffer.py",
 ])
 def test_script_help(script, tmp_path):
     r = _run(script, "--help", cwd=tmp_path)
@@ -122,3 +123,19 @@ def test_analysis_cli(tmp_path):
     assert (tmp_path / "influence.npz").exists()
     img = np.load(tmp_path / "inf.npy")
     assert img.shape == (16, 16) and np.isfinite(img).all()
+
+
+def test_inspect_replaybuffer(tmp_path):
+    import torch
+    sys.path.insert(0, str(ROOT))
+    from smartcal_amd.rl.buffers_dict import DictReplayBuffer
+    b = DictReplayBuffer(32, (1, 16, 16), (4,), 2)
+    s0 = {"infmap": torch.rand(1, 16, 16), "metadata": torch.rand(4)}
+    for _ in range(5):
+        b.store_transition(s0, torch.rand(2), 0.1, s0, False,
+                           torch.zeros(2))
+    b.save_checkpoint(str(tmp_path / "buf.pkl"))
+    r = _run("scripts/calibration/inspect_replaybuffer.py", "buf.pkl",
+             "--out", "rb.png", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (tmp_path / "rb.png").exists()
