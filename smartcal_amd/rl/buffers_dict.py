@@ -128,11 +128,12 @@ class DictPERBuffer(DictReplayBuffer):
     MAX_PRIORITY = 1.0
 
     def __init__(self, max_size, img_shape, meta_shape, n_actions,
-                 device=None):
+                 device=None, normalize_reward: bool = False):
         super().__init__(max_size, img_shape, meta_shape, n_actions, device)
         self.priorities = torch.zeros(self.mem_size, dtype=torch.float32,
                                       device=self.device)
         self.beta = self.BETA0
+        self.normalize_reward = normalize_reward
         self.filename = "prioritized_replaymem_cnn.model"
 
     def store_transition(self, state, action, reward, state_, done,
@@ -148,7 +149,14 @@ class DictPERBuffer(DictReplayBuffer):
                                                batch_size)
         self.beta = min(1.0, self.beta + self.BETA_INC)
         weights = per_ops.importance_weights(probs, n, self.beta)
-        return self._gather(idx), idx, weights
+        batch = self._gather(idx)
+        if self.normalize_reward:
+            # reference `demix_td3.py:162-166`: standardize sampled rewards
+            # by the filled buffer's running statistics
+            r = self.reward_memory[:n]
+            mu, sd = r.mean(), r.std().clamp(min=1e-6)
+            batch = (*batch[:3], (batch[3] - mu) / sd, *batch[4:])
+        return batch, idx, weights
 
     def update_priorities(self, idx, td_errors):
         pri = (td_errors.detach().abs().reshape(-1) + self.EPS) \

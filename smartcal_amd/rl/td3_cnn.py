@@ -29,6 +29,7 @@ class Agent:
                  max_mem_size=100, tau=0.005, M=3, meta_dim=None,
                  update_actor_interval=2, warmup=100, noise=0.1,
                  name_prefix="", use_hint=False, prioritized=True,
+                 normalize_reward=False,
                  admm_rho=0.1, device: Optional[torch.device] = None,
                  checkpoint_dir="./", grad_hook=None):
         self.gamma = gamma
@@ -53,9 +54,17 @@ class Agent:
             meta_dim = 7 * (M + 1)
         self.meta_dim = meta_dim
         self.prioritized = prioritized
-        buf_cls = DictPERBuffer if prioritized else DictReplayBuffer
-        self.replaymem = buf_cls(max_mem_size, img_shape, meta_dim,
-                                 n_actions, device=self.device)
+        if prioritized:
+            # reference `demix_td3.py:162-166,380`: PER with optional
+            # reward standardization over the stored buffer
+            self.replaymem = DictPERBuffer(max_mem_size, img_shape,
+                                           meta_dim, n_actions,
+                                           device=self.device,
+                                           normalize_reward=normalize_reward)
+        else:
+            self.replaymem = DictReplayBuffer(max_mem_size, img_shape,
+                                              meta_dim, n_actions,
+                                              device=self.device)
 
         mk_actor = lambda: DeterministicActorCNN(hw, meta_dim, n_actions)
         mk_critic = lambda: CriticCNN(hw, meta_dim, n_actions)

@@ -102,3 +102,19 @@ def test_agent_on_env_obs():
     for _ in range(4):
         agent.store_transition(obs, a, r, obs2, done)
     agent.learn()
+
+
+def test_per_normalize_reward():
+    """normalize_reward standardizes sampled rewards (`demix_td3.py:162`)."""
+    import torch
+    from smartcal_amd.rl.buffers_dict import DictPERBuffer
+    b = DictPERBuffer(64, (1, 8, 8), (3,), 2, normalize_reward=True)
+    s = {"infmap": torch.rand(1, 8, 8), "metadata": torch.rand(3)}
+    for i in range(32):
+        b.store_transition(s, torch.rand(2), float(i), s, False,
+                           torch.zeros(2))
+    batch, idx, w = b.sample_buffer(16)
+    r = batch[3]
+    assert r.abs().max() < 4.0          # standardized scale
+    raw = b.reward_memory[:32]
+    assert raw.max() == 31.0            # stored rewards untouched
