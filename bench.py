@@ -102,12 +102,23 @@ def main():
     if have_gpu:
         # capture the whole learn step into one hipGraph (forwards,
         # backwards, fused Adam, polyak — single replay per step)
+        graph_ok = True
         try:
             agent.enable_cuda_graph()
         except Exception as e:  # noqa: BLE001
             import sys as _sys
+            graph_ok = False
             print(f"[bench] graph capture unavailable, eager path: {e}",
                   file=_sys.stderr)
+        if world > 1:
+            # all ranks must agree on graphed vs eager: capture records
+            # (does not execute) the in-graph all_reduce, so a rank that
+            # replays while another runs eager would desequence RCCL —
+            # take the AND over ranks and fall back together
+            ok = torch.tensor([1.0 if graph_ok else 0.0], device=device)
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if ok.item() < 1.0 and graph_ok:
+                agent._graph = None
 
     # ---- warmup ----
     for _ in range(args.warmup):

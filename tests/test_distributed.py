@@ -218,3 +218,25 @@ def test_learner_tolerates_actor_env_failure():
     scores = q.get()
     # 3 episodes ran; the failed round contributed no transitions
     assert len(scores) >= 2
+
+
+def test_bench_torchrun_world2_cpu(tmp_path):
+    """The driver's exact multi-rank launch contract: torchrun world=2
+    over gloo on CPU must print one valid JSON line from rank 0."""
+    import json as _json
+    import subprocess, os, sys
+    from pathlib import Path
+    ROOT = Path(__file__).resolve().parents[1]
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", str(ROOT / "bench.py"),
+         "--gpus", "2", "--steps", "4", "--warmup", "1"],
+        cwd=tmp_path, env=env, capture_output=True, text=True, timeout=540)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = _json.loads(line)
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["value"] > 0
