@@ -98,6 +98,12 @@ class ENetEnv(gymapi.Env):
             action = torch.as_tensor(np.asarray(action).reshape(-1),
                                      dtype=torch.float32)
         action = action.detach().reshape(-1)
+        # the reference stack never emits NaN (lbfgsnew's guards +
+        # torch.linalg.eig); keep that contract at the env boundary so a
+        # rare degenerate solve can't poison the replay buffer. Finite
+        # extremes pass through untouched (reference reward is unclamped,
+        # `enetenv.py:149-150`).
+        action = torch.nan_to_num(action, nan=0.0, posinf=1.0, neginf=-1.0)
 
         if self.device.type == "cuda" and action.device == self.device:
             # fully device-resident step: no host round trip anywhere
@@ -110,6 +116,9 @@ class ENetEnv(gymapi.Env):
             x, EE, reward = enet_ops.solve_and_influence_device(
                 self.A, self.y, rho, penalty.to(torch.float32))
             self.x = x
+            EE = torch.nan_to_num(EE, nan=0.0, posinf=1e6, neginf=-1e6)
+            reward = torch.nan_to_num(reward, nan=-100.0, posinf=1e6,
+                                      neginf=-1e6)
             observation = {"A": self.A.reshape(-1), "eig": EE}
             info: dict = {}
             if self.provide_hint:
@@ -136,6 +145,10 @@ class ENetEnv(gymapi.Env):
         x, EE, reward = enet_ops.solve_and_influence(
             self.A, self.y, float(self.rho[0]), float(self.rho[1]), penalty)
         self.x = x
+        EE = torch.nan_to_num(EE, nan=0.0, posinf=1e6, neginf=-1e6)
+        reward = torch.nan_to_num(torch.as_tensor(reward,
+                                                  dtype=torch.float32),
+                                  nan=-100.0, posinf=1e6, neginf=-1e6)
 
         observation = {"A": self.A.reshape(-1), "eig": EE}
         info: dict = {}

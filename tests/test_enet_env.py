@@ -113,3 +113,23 @@ def test_skenet_gridsearchcv():
     assert gs.best_score_ > -0.01
     est = SKEnet(**gs.best_params_).fit(A, y)
     assert np.mean((est.predict(A) - y) ** 2) < 1e-3
+
+
+def test_env_sanitizes_nonfinite(monkeypatch):
+    """NaN from a degenerate solve must not escape the env boundary."""
+    import torch
+    import numpy as np
+    from smartcal_amd.envs import enet as enet_env_mod
+    env = enet_env_mod.ENetEnv(8, 8)
+    env.reset()
+
+    def bad_solve(A, y, r1, r2, pen):
+        x = torch.full((8,), float("nan"))
+        EE = torch.full((8,), float("nan"))
+        return x, EE, float("nan")
+
+    monkeypatch.setattr(enet_env_mod.enet_ops, "solve_and_influence",
+                        bad_solve)
+    obs, r, done, info = env.step(np.array([0.5, float("nan")]))
+    assert np.isfinite(r)
+    assert torch.isfinite(obs["eig"]).all()
