@@ -118,7 +118,7 @@ def main():
             ok = torch.tensor([1.0 if graph_ok else 0.0], device=device)
             dist.all_reduce(ok, op=dist.ReduceOp.MIN)
             if ok.item() < 1.0 and graph_ok:
-                agent._graph = None
+                agent.disable_cuda_graph()
 
     # ---- warmup ----
     for _ in range(args.warmup):

@@ -300,6 +300,11 @@ class Agent:
                              self._g_done, self._g_hint)
         self._graph = g
 
+    def disable_cuda_graph(self):
+        """Drop a captured graph and return to the eager learn path
+        (used e.g. when ranks must agree on graphed vs eager mode)."""
+        self._graph = None
+
     def _learn_graphed(self):
         mem = self.replaymem
         n = len(mem)
