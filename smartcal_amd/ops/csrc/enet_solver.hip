@@ -506,14 +506,22 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
   for (int m = 0; m < 32; ++m)
     qc[m] = (lane < N && m < M) ? -2.f * A[lane * M + m] : 0.f;
 
+  // Degenerate-pair filter: the true Hessian is 2(A^T A + rho1 I), so
+  // ys/ss >= 2 rho1 for any genuine curvature pair; pairs far below that
+  // are line-search noise whose 1/ys blows the two-loop up (observed as
+  // ~1e6-magnitude eigenvalue-ratio rewards). ro = 0 no-ops a pair in
+  // both loops. The CPU oracle (`ops/enet.py::influence_eigs_reference`)
+  // applies the same filter.
   float ys = 1.f, yy = 1.f;
   for (int i = 0; i < nh; ++i) {
-    float r = 1.f / lds_dot(&Yv[i * M], &Sv[i * M], M);
+    float ysi = lds_dot(&Yv[i * M], &Sv[i * M], M);
+    float ssi = lds_dot(&Sv[i * M], &Sv[i * M], M);
+    float r = (ysi > 1e-6f * ssi) ? 1.f / ysi : 0.f;
     if (lane == 0) ro[i] = r;
-  }
-  if (nh > 0) {
-    ys = lds_dot(&Yv[(nh - 1) * M], &Sv[(nh - 1) * M], M);
-    yy = lds_dot(&Yv[(nh - 1) * M], &Yv[(nh - 1) * M], M);
+    if (r != 0.f) {  // H_diag scale from the newest GOOD pair
+      ys = ysi;
+      yy = lds_dot(&Yv[i * M], &Yv[i * M], M);
+    }
   }
   __builtin_amdgcn_s_barrier();
 
@@ -536,7 +544,7 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
 #pragma unroll
     for (int m = 0; m < 32; ++m) qc[m] -= yv[m] * ai;
   }
-  const float scale = (nh > 0) ? (ys / yy) : 1.f;
+  const float scale = ys / yy;  // 1 when no good pair (init values)
 #pragma unroll
   for (int m = 0; m < 32; ++m) qc[m] *= scale;
 #pragma unroll

@@ -81,6 +81,15 @@ def influence_eigs_reference(A: torch.Tensor, Y: torch.Tensor,
     """
     from ..autograd_tools import inv_hessian_mult_mat
 
+    if Y.shape[0] > 0:
+        # degenerate-pair filter (same as the HIP influence kernel): the
+        # true Hessian 2(A^T A + rho1 I) bounds ys/ss >= 2 rho1, so pairs
+        # far below are line-search noise whose 1/ys explodes the
+        # two-loop (and with it the min(EE)/max(EE) reward term)
+        ys = (Y * S).sum(-1)
+        ss = (S * S).sum(-1)
+        good = ys > 1e-6 * ss
+        Y, S = Y[good], S[good]
     Q = -2.0 * A.t().contiguous()
     mm = inv_hessian_mult_mat(Y, S, Q)
     B = A @ mm

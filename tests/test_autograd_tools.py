@@ -75,3 +75,23 @@ def test_influence_matrix_runs():
     If = at.influence_matrix(net, x, y)
     assert If.shape == (2, 3)
     assert torch.isfinite(If).all()
+
+
+def test_influence_degenerate_pair_filter():
+    """Near-zero-curvature pairs must not explode the influence eigens.
+
+    The true Hessian 2(A^T A + rho1 I) bounds ys/ss >= 2 rho1, so a pair
+    with ys/ss ~ 1e-7 is line-search noise; unfiltered it drives
+    min(EE)/max(EE) to ~1e4+ (the -4e6 rewards seen in the hint arm).
+    """
+    import torch
+    from smartcal_amd.ops import enet as enet_ops
+    torch.manual_seed(0)
+    A = torch.randn(20, 20)
+    A = A / A.norm()
+    s = torch.randn(20) * 1e-4
+    y = s * 1e-6 + torch.randn(20) * 1e-10
+    EE = enet_ops.influence_eigs_reference(A, y.unsqueeze(0),
+                                           s.unsqueeze(0))
+    ratio = float(EE.min() / EE.max())
+    assert -2.0 <= ratio <= 1.0
