@@ -23,7 +23,7 @@ solve is a handful of GPU ops instead of N × history python-loop round trips.
 
 from __future__ import annotations
 
-from typing import List, Sequence, Tuple
+from typing import Sequence, Tuple
 
 import torch
 

@@ -21,7 +21,6 @@ Reward:  σ_data/σ_res + 1e-4/(σ_inf+EPS) + bound-penalty
 
 from __future__ import annotations
 
-import math
 
 import numpy as np
 import torch

@@ -15,7 +15,6 @@ index — where the reference rewrites cluster text files and re-runs
 from __future__ import annotations
 
 import itertools
-import math
 
 import numpy as np
 import torch

@@ -11,7 +11,6 @@ inverse FFT; conjugate symmetry is enforced so images are real.
 
 from __future__ import annotations
 
-import math
 
 import torch
 

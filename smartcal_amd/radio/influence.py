@@ -17,7 +17,7 @@ import torch
 
 from . import hessian as hs
 from .consensus import hessian_addition_scalar
-from .sim import VisData, to_R
+from .sim import to_R
 
 __all__ = ["influence_values", "influence_per_direction", "hadd_for"]
 

@@ -19,7 +19,7 @@ import numpy as np
 import torch
 
 from . import array as arr
-from .coords import radectolm, lmtoradec
+from .coords import lmtoradec
 from .sky import SkyModel, ClusterSet, ClusterDef
 from .coherency import predict_coherencies_uvw
 from .solutions import simulate_systematic_errors, solutions_to_J

@@ -27,7 +27,6 @@ pipeline (`radio.influence`) and text writers consume them directly.
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 
 import numpy as np
