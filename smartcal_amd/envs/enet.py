@@ -227,10 +227,16 @@ def obs_to_state(observation) -> torch.Tensor:
         A = torch.as_tensor(A, dtype=torch.float32)
     return torch.cat((eig.reshape(-1), A.reshape(-1).to(eig.device)))
 
-class SKEnet:
+try:
+    from sklearn.base import BaseEstimator as _SKBase
+except Exception:  # pragma: no cover - sklearn always present in-image
+    _SKBase = object
+
+
+class SKEnet(_SKBase):
     """sklearn-compatible elastic-net estimator (reference
-    ``elasticnet/enetenv.py:249-295``): exposes fit/predict/score plus
-    get_params/set_params so it plugs into ``sklearn.model_selection
+    ``elasticnet/enetenv.py:249-295``): a ``BaseEstimator`` with
+    fit/predict/score so it plugs into ``sklearn.model_selection
     .GridSearchCV`` exactly like the reference's wrapper. The fit itself
     runs the in-framework L-BFGS solver (CPU reference path, so it works
     without a GPU) instead of scipy L-BFGS-B."""
@@ -239,14 +245,6 @@ class SKEnet:
         self.lambda1 = lambda1  # L1 weight
         self.lambda2 = lambda2  # L2 weight
         self.x_ = None
-
-    def get_params(self, deep=True):
-        return {"lambda1": self.lambda1, "lambda2": self.lambda2}
-
-    def set_params(self, **params):
-        for k, v in params.items():
-            setattr(self, k, v)
-        return self
 
     def fit(self, A, y):
         A = torch.as_tensor(np.asarray(A), dtype=torch.float32)

@@ -131,5 +131,5 @@ def test_env_sanitizes_nonfinite(monkeypatch):
     monkeypatch.setattr(enet_env_mod.enet_ops, "solve_and_influence",
                         bad_solve)
     obs, r, done, info = env.step(np.array([0.5, float("nan")]))
-    assert np.isfinite(r)
+    assert np.isfinite(float(r))
     assert torch.isfinite(obs["eig"]).all()
