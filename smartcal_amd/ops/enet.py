@@ -143,3 +143,36 @@ def solve_and_influence(A: torch.Tensor, y: torch.Tensor,
     final_err = torch.norm(A @ x - y, 2)
     reward = torch.norm(y, 2) / final_err + EE.min() / EE.max() + penalty
     return x, EE, reward
+
+
+def solve_and_influence_batch(A: torch.Tensor, y: torch.Tensor,
+                              rho: torch.Tensor, penalty: torch.Tensor,
+                              epochs: int = 20, max_iter: int = 10,
+                              history: int = 7):
+    """Batched env-step compute over E independent problems.
+
+    A (E, N, M), y (E, N), rho (E, 2), penalty (E,) → x (E, M),
+    EE (E, N), reward (E,). On GPU this is exactly the vectorized-env
+    layout the kernels are built for: one workgroup per environment
+    (``enet_solver.hip`` blockIdx.x), E solves in one launch. The CPU
+    path loops the reference oracle.
+    """
+    from . import use_hip
+
+    if use_hip(A):
+        Ab = A.contiguous()
+        yb = y.contiguous()
+        x, Yc, Sc, nh = ext().enet_lbfgs_solve(Ab, yb, rho.contiguous(),
+                                               epochs, max_iter, history)
+        EE, reward = ext().enet_influence(Ab, yb, x, Yc, Sc, nh,
+                                          penalty.contiguous())
+        return x, EE, reward
+    xs, EEs, rs = [], [], []
+    for e in range(A.shape[0]):
+        xe, EEe, re = solve_and_influence(
+            A[e], y[e], float(rho[e, 0]), float(rho[e, 1]),
+            float(penalty[e]), epochs, max_iter, history)
+        xs.append(xe)
+        EEs.append(EEe)
+        rs.append(torch.as_tensor(re, dtype=torch.float32))
+    return torch.stack(xs), torch.stack(EEs), torch.stack(rs)

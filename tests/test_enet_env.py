@@ -133,3 +133,25 @@ def test_env_sanitizes_nonfinite(monkeypatch):
     obs, r, done, info = env.step(np.array([0.5, float("nan")]))
     assert np.isfinite(float(r))
     assert torch.isfinite(obs["eig"]).all()
+
+
+def test_vec_enet_env_contract():
+    """VecENetEnv: batched shapes, finite rewards, per-env independence."""
+    import torch
+    from smartcal_amd.envs.vec_enet import VecENetEnv
+    torch.manual_seed(0)
+    np.random.seed(0)
+    env = VecENetEnv(3, 8, 8)
+    obs = env.reset()
+    assert obs["A"].shape == (3, 64) and obs["eig"].shape == (3, 8)
+    a = np.random.uniform(-1, 1, size=(3, 2)).astype(np.float32)
+    obs2, r, done, info = env.step(a)
+    assert r.shape == (3,) and torch.isfinite(r).all()
+    assert obs2["eig"].shape == (3, 8)
+    assert not done.any()
+    err = env.solution_error()
+    assert err.shape == (3,) and torch.isfinite(err).all()
+    # out-of-range actions get the boundary penalty per env
+    obs3, r3, *_ = env.step(np.array([[2.0, 2.0], [0.0, 0.0],
+                                      [-2.0, -2.0]], dtype=np.float32))
+    assert torch.isfinite(r3).all()
