@@ -91,6 +91,7 @@ class Agent:
                  alpha=0.1, hint_threshold=0.1, admm_rho=1.0,
                  name_prefix="", use_hint=False, prioritized=False,
                  use_influence=True, meta_dim=None, learn_alpha=False,
+                 arch="cnn",
                  device: Optional[torch.device] = None,
                  checkpoint_dir="./", grad_hook=None):
         self.gamma = gamma
@@ -115,7 +116,16 @@ class Agent:
         self.replaymem = buf_cls(max_mem_size, img_shape, meta_dim,
                                  n_actions, device=self.device)
 
-        if use_influence:
+        if use_influence and arch == "transformer":
+            # BASELINE.json config: "calibenv SAC with transformer
+            # actor/critic" — token-sequence encoder over the sky
+            # metadata rows + influence-map embedding
+            from .transformer_networks import (SACActorTransformer,
+                                               TransformerCritic)
+            mk_actor = lambda: SACActorTransformer(hw, meta_dim, n_actions,
+                                                   self.max_action)
+            mk_critic = lambda: TransformerCritic(hw, meta_dim, n_actions)
+        elif use_influence:
             mk_actor = lambda: SACActorCNN(hw, meta_dim, n_actions,
                                            self.max_action)
             mk_critic = lambda: CriticCNN(hw, meta_dim, n_actions)

@@ -118,3 +118,28 @@ def test_per_normalize_reward():
     assert r.abs().max() < 4.0          # standardized scale
     raw = b.reward_memory[:32]
     assert raw.max() == 31.0            # stored rewards untouched
+
+
+def test_transformer_agent_learn():
+    """SAC with transformer actor/critic (BASELINE.json calibenv config):
+    action, store, one learn step on CPU."""
+    import torch
+    from smartcal_amd.rl.sac_cnn import Agent
+    torch.manual_seed(0)
+    M = 3
+    agent = Agent(gamma=0.99, batch_size=4, n_actions=2 * M, tau=0.005,
+                  max_mem_size=32, input_dims=(1, 32, 32), M=M,
+                  lr_a=1e-3, lr_c=1e-3, arch="transformer")
+    from smartcal_amd.rl.transformer_networks import (SACActorTransformer,
+                                                      TransformerCritic)
+    assert isinstance(agent.actor, SACActorTransformer)
+    assert isinstance(agent.critic_1, TransformerCritic)
+    s = {"img": torch.rand(1, 32, 32), "sky": torch.rand(7 * (M + 1))}
+    a = agent.choose_action(s)
+    assert np.asarray(a).shape == (2 * M,)
+    for _ in range(6):
+        agent.store_transition(s, a, 0.5, s, False,
+                               np.zeros(2 * M, np.float32))
+    agent.learn()
+    a2 = agent.choose_action(s)
+    assert np.isfinite(np.asarray(a2)).all()

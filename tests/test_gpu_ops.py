@@ -404,3 +404,25 @@ def test_vec_enet_env_gpu():
     obs, r, done, _ = env.step(a)
     assert r.shape == (16,) and torch.isfinite(r).all()
     assert obs["eig"].shape == (16, 20)
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason='needs GPU')
+def test_transformer_agent_learn_gpu():
+    """Transformer actor/critic SAC learn step on the fused-linear HIP
+    kernels (BASELINE.json calibenv transformer config)."""
+    from smartcal_amd.rl.sac_cnn import Agent
+    torch.manual_seed(0)
+    M = 3
+    agent = Agent(gamma=0.99, batch_size=4, n_actions=2 * M, tau=0.005,
+                  max_mem_size=32, input_dims=(1, 32, 32), M=M,
+                  lr_a=1e-3, lr_c=1e-3, arch="transformer",
+                  device=torch.device("cuda"))
+    s = {"img": torch.rand(1, 32, 32), "sky": torch.rand(7 * (M + 1))}
+    a = agent.choose_action(s)
+    for _ in range(6):
+        agent.store_transition(s, a, 0.5, s, False,
+                               np.zeros(2 * M, np.float32))
+    agent.learn()
+    torch.cuda.synchronize()
+    a2 = np.asarray(agent.choose_action(s))
+    assert np.isfinite(a2).all()

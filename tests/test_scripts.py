@@ -51,10 +51,11 @@ def test_enet_sac_short(tmp_path):
     assert (tmp_path / "scores.pkl").exists()
 
 
-def test_calib_sac_short(tmp_path):
+@pytest.mark.parametrize("arch", ["cnn", "transformer"])
+def test_calib_sac_short(tmp_path, arch):
     r = _run("scripts/calibration/main_sac.py", "--episodes", "1",
              "--steps", "1", "--M", "3", "--stations", "8",
-             cwd=tmp_path)
+             "--arch", arch, cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
 
 
