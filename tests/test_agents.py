@@ -134,3 +134,32 @@ def test_seeded_determinism():
     s1 = run()
     s2 = run()
     np.testing.assert_allclose(s1, s2, rtol=0, atol=0)
+
+
+def test_checkpoint_file_layout(tmp_path):
+    """The saved-model layout matches the reference's filenames exactly
+    (`enet_sac.py:396-403`, `enet_td3.py:151-159`, `enet_ddpg.py`), so a
+    reference user's resume scripts keep working."""
+    import os
+    from smartcal_amd.rl.sac import Agent as SAC
+    from smartcal_amd.rl.td3 import Agent as TD3
+    from smartcal_amd.rl.ddpg import Agent as DDPG
+    kw = dict(gamma=0.99, batch_size=4, n_actions=2, tau=0.005,
+              max_mem_size=16, input_dims=[12], lr_a=1e-3, lr_c=1e-3,
+              checkpoint_dir=str(tmp_path))
+    SAC(reward_scale=2, alpha=0.03, **kw).save_models()
+    TD3(**kw).save_models()
+    DDPG(**kw).save_models()
+    files = set(os.listdir(tmp_path))
+    expect = {
+        "a_eval_sac_actor.model", "q_eval_1_sac_critic.model",
+        "q_eval_2_sac_critic.model", "replaymem_sac.model",
+        "a_eval_td3_actor.model", "a_target_td3_actor.model",
+        "q_eval_1_td3_critic.model", "q_eval_2_td3_critic.model",
+        "q_target_1_td3_critic.model", "q_target_2_td3_critic.model",
+        "a_eval_ddpg_actor.model", "a_target_ddpg_actor.model",
+        "q_eval_ddpg_critic.model", "q_target_ddpg_critic.model",
+        "replaymem_ddpg.model",
+    }
+    missing = expect - files
+    assert not missing, f"missing reference-layout files: {missing}"
