@@ -37,6 +37,7 @@ def _run(script, *args, cwd):
     "scripts/elasticnet/distributed_per_sac.py",
     "scripts/demixing_rl/distributed_per_sac.py",
     "scripts/calibration/inspect_replaybuffer.py",
+    "scripts/elasticnet/enet_eval.py",
 ])
 def test_script_help(script, tmp_path):
     r = _run(script, "--help", cwd=tmp_path)
@@ -49,6 +50,10 @@ def test_enet_sac_short(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     assert "episode 1" in r.stdout
     assert (tmp_path / "scores.pkl").exists()
+    # eval CLI consumes the checkpoints just written (RL vs GridSearchCV)
+    r = _run("scripts/elasticnet/enet_eval.py", "--episodes", "1",
+             cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
 
 
 @pytest.mark.parametrize("arch", ["cnn", "transformer"])
