@@ -240,3 +240,30 @@ def test_bench_torchrun_world2_cpu(tmp_path):
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"] == "dp2"
     assert out["value"] > 0
+
+
+def test_demix_flat_obs_agent_split():
+    """The demixing distributed CLI ingests flat records and must split
+    them back into {infmap, metadata} for the CNN buffer."""
+    import importlib.util
+    from pathlib import Path
+    ROOT = Path(__file__).resolve().parents[1]
+    spec = importlib.util.spec_from_file_location(
+        "dps_demix", ROOT / "scripts" / "demixing_rl"
+        / "distributed_per_sac.py")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    agent = mod._FlatObsAgent(
+        gamma=0.99, batch_size=2, n_actions=mod.K, tau=0.005,
+        max_mem_size=8, input_dims=(1, mod.NINF, mod.NINF),
+        meta_dim=mod.META, lr_a=3e-4, lr_c=3e-4, prioritized=True,
+        use_hint=True)
+    flat = np.arange(mod.OBS_DIM, dtype=np.float32)
+    agent.store_transition(flat, np.zeros(mod.K, np.float32), 1.0,
+                           flat, False, np.zeros(mod.K, np.float32))
+    img = agent.replaymem.img_memory[0]
+    meta = agent.replaymem.meta_memory[0]
+    assert float(img.reshape(-1)[0]) == 0.0
+    assert float(img.reshape(-1)[-1]) == mod.NINF * mod.NINF - 1
+    assert float(meta[0]) == mod.NINF * mod.NINF
+    assert float(meta[-1]) == mod.OBS_DIM - 1
