@@ -24,7 +24,9 @@ def _need_gpu():
 
 def test_fused_linear_fwd_matches_cpu():
     for (B, K, N) in [(64, 420, 512), (1, 420, 512), (64, 512, 256),
-                      (64, 128, 4), (3, 33, 17), (64, 320, 1)]:
+                      (64, 128, 4), (3, 33, 17), (64, 320, 1),
+                      (4, 7, 64), (20, 7, 64), (5, 2, 192),
+                      (4, 64, 192)]:
         x = torch.randn(B, K)
         W = torch.randn(N, K) * 0.05
         b = torch.randn(N) * 0.1
@@ -42,9 +44,10 @@ def test_fused_linear_fwd_matches_cpu():
         assert torch.allclose(y2.cpu(), ref2, atol=2e-4)
 
 
-def test_fused_linear_backward_matches_cpu():
+@pytest.mark.parametrize("B,K,N", [(64, 420, 512), (4, 7, 64),
+                                   (5, 2, 192)])
+def test_fused_linear_backward_matches_cpu(B, K, N):
     torch.manual_seed(0)
-    B, K, N = 64, 420, 512
     x = torch.randn(B, K)
     W = torch.randn(N, K) * 0.05
     b = torch.randn(N) * 0.1
