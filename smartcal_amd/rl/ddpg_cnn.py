@@ -24,7 +24,8 @@ from .noise import OUActionNoise
 class Agent:
     def __init__(self, gamma, lr_a, lr_c, input_dims, batch_size, n_actions,
                  max_mem_size=100, tau=0.001, M=3, meta_dim=None,
-                 name_prefix="", device: Optional[torch.device] = None,
+                 name_prefix="", arch="cnn",
+                 device: Optional[torch.device] = None,
                  checkpoint_dir="./", grad_hook=None):
         self.gamma = gamma
         self.tau = tau
@@ -44,8 +45,15 @@ class Agent:
                                           n_actions, device=self.device)
         self.noise = OUActionNoise(mu=np.zeros(n_actions))
 
-        mk_actor = lambda: DeterministicActorCNN(hw, meta_dim, n_actions)
-        mk_critic = lambda: CriticCNN(hw, meta_dim, n_actions)
+        if arch == "transformer":
+            from .transformer_networks import (DeterministicActorTransformer,
+                                               TransformerCritic)
+            mk_actor = lambda: DeterministicActorTransformer(hw, meta_dim,
+                                                             n_actions)
+            mk_critic = lambda: TransformerCritic(hw, meta_dim, n_actions)
+        else:
+            mk_actor = lambda: DeterministicActorCNN(hw, meta_dim, n_actions)
+            mk_critic = lambda: CriticCNN(hw, meta_dim, n_actions)
         self.actor = mk_actor().to(self.device)
         self.target_actor = mk_actor().to(self.device)
         self.critic = mk_critic().to(self.device)

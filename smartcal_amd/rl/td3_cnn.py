@@ -29,7 +29,7 @@ class Agent:
                  max_mem_size=100, tau=0.005, M=3, meta_dim=None,
                  update_actor_interval=2, warmup=100, noise=0.1,
                  name_prefix="", use_hint=False, prioritized=True,
-                 normalize_reward=False,
+                 normalize_reward=False, arch="cnn",
                  admm_rho=0.1, device: Optional[torch.device] = None,
                  checkpoint_dir="./", grad_hook=None):
         self.gamma = gamma
@@ -66,8 +66,15 @@ class Agent:
                                               meta_dim, n_actions,
                                               device=self.device)
 
-        mk_actor = lambda: DeterministicActorCNN(hw, meta_dim, n_actions)
-        mk_critic = lambda: CriticCNN(hw, meta_dim, n_actions)
+        if arch == "transformer":
+            from .transformer_networks import (DeterministicActorTransformer,
+                                               TransformerCritic)
+            mk_actor = lambda: DeterministicActorTransformer(hw, meta_dim,
+                                                             n_actions)
+            mk_critic = lambda: TransformerCritic(hw, meta_dim, n_actions)
+        else:
+            mk_actor = lambda: DeterministicActorCNN(hw, meta_dim, n_actions)
+            mk_critic = lambda: CriticCNN(hw, meta_dim, n_actions)
         self.actor = mk_actor().to(self.device)
         self.target_actor = mk_actor().to(self.device)
         self.critic_1 = mk_critic().to(self.device)

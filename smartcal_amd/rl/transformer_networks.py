@@ -165,3 +165,19 @@ class SACActorTransformer(nn.Module):
         log_probs = dist.log_prob(raw) \
             - torch.log(1 - action.pow(2) + self.reparam_noise)
         return action, log_probs.sum(1, keepdim=True)
+
+
+class DeterministicActorTransformer(nn.Module):
+    """tanh deterministic policy with the token-transformer trunk
+    (TD3/DDPG counterpart of :class:`SACActorTransformer`)."""
+
+    def __init__(self, img_hw, meta_dim: int, n_actions: int,
+                 d_model: int = 64):
+        super().__init__()
+        self.trunk = _ObsEncoder(img_hw, meta_dim, 0, d_model=d_model)
+        self.out = nn.Linear(d_model, n_actions)
+        nn.init.uniform_(self.out.weight, -3e-3, 3e-3)
+        nn.init.zeros_(self.out.bias)
+
+    def forward(self, img, meta):
+        return torch.tanh(self.out(self.trunk(img, meta)))

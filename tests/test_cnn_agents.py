@@ -167,3 +167,28 @@ def test_transformer_networks_trainable():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < 0.3 * losses[0], (losses[0], losses[-1])
+
+
+@pytest.mark.parametrize("mod", ["td3_cnn", "ddpg_cnn"])
+def test_transformer_arch_td3_ddpg(mod):
+    """TD3/DDPG calib agents accept arch='transformer' and learn a step."""
+    import importlib
+    import torch
+    m = importlib.import_module(f"smartcal_amd.rl.{mod}")
+    torch.manual_seed(0)
+    M = 3
+    agent = m.Agent(gamma=0.99, batch_size=4, n_actions=2 * M, tau=0.005,
+                    max_mem_size=32, input_dims=(1, 32, 32), M=M,
+                    lr_a=1e-3, lr_c=1e-3, arch="transformer")
+    from smartcal_amd.rl.transformer_networks import (
+        DeterministicActorTransformer, TransformerCritic)
+    assert isinstance(agent.actor, DeterministicActorTransformer)
+    s = {"img": torch.rand(1, 32, 32), "sky": torch.rand(7 * (M + 1))}
+    a = agent.choose_action(s)
+    for _ in range(6):
+        agent.store_transition(s, a, 0.5, s, False,
+                               np.zeros(2 * M, np.float32))
+    agent.learn()
+    agent.learn()
+    a2 = agent.choose_action(s)
+    assert np.isfinite(np.asarray(a2)).all()
