@@ -21,12 +21,14 @@ def main():
     ap.add_argument("--steps", default=4, type=int)
     ap.add_argument("--M", default=10, type=int)
     ap.add_argument("--stations", default=62, type=int)
+    ap.add_argument("--arch", default="cnn", choices=("cnn", "transformer"))
     args = ap.parse_args()
     seed_everything(args.seed)
 
     M = args.M
     env = CalibEnv(M=M, N_stations=args.stations, seed=args.seed)
-    agent = Agent(gamma=0.99, batch_size=32, n_actions=2 * M, tau=0.001,
+    agent = Agent(arch=args.arch,
+                  gamma=0.99, batch_size=32, n_actions=2 * M, tau=0.001,
                   max_mem_size=10000, input_dims=(1, 128, 128), M=M,
                   lr_a=1e-3, lr_c=1e-3)
     run_training(env, agent, args.episodes, args.steps)

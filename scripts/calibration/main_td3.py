@@ -22,13 +22,15 @@ def main():
     ap.add_argument("--use_hint", action="store_true", default=False)
     ap.add_argument("--M", default=10, type=int)
     ap.add_argument("--stations", default=62, type=int)
+    ap.add_argument("--arch", default="cnn", choices=("cnn", "transformer"))
     args = ap.parse_args()
     seed_everything(args.seed)
 
     M = args.M
     env = CalibEnv(M=M, provide_hint=args.use_hint,
                    N_stations=args.stations, seed=args.seed)
-    agent = Agent(gamma=0.99, batch_size=32, n_actions=2 * M, tau=0.005,
+    agent = Agent(arch=args.arch,
+                  gamma=0.99, batch_size=32, n_actions=2 * M, tau=0.005,
                   max_mem_size=10000, input_dims=(1, 128, 128), M=M,
                   lr_a=1e-3, lr_c=1e-3, warmup=100, noise=0.1,
                   prioritized=True, use_hint=args.use_hint)
