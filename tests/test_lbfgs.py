@@ -149,3 +149,32 @@ def test_batch_mode_descends():
     for _ in range(30):
         opt.step(closure)
     assert torch.norm(x.detach() - xt) < 0.5
+
+
+def test_enet_solve_matches_scipy_lbfgsb():
+    """Our L-BFGS elastic-net solve reaches the same objective as scipy
+    L-BFGS-B (the reference's SKEnet inner solver, `enetenv.py:249-295`)
+    on random instances."""
+    import numpy as np
+    import torch
+    from scipy.optimize import minimize
+    from smartcal_amd.ops import enet as enet_ops
+
+    rng = np.random.default_rng(0)
+    for _ in range(3):
+        N = M = 12
+        A = rng.normal(size=(N, M)).astype(np.float32)
+        A /= np.linalg.norm(A)
+        y = (A @ (rng.normal(size=M) * (rng.random(M) > 0.6))) \
+            .astype(np.float32)
+        rho1, rho2 = 0.05, 0.02
+
+        def f(x):
+            r = y - A @ x
+            return float(r @ r + rho1 * x @ x + rho2 * np.abs(x).sum())
+
+        sp = minimize(f, np.zeros(M), method="L-BFGS-B")
+        x_t, _ = enet_ops.lbfgs_solve_reference(
+            torch.from_numpy(A), torch.from_numpy(y), rho1, rho2)
+        ours = f(x_t.numpy().astype(np.float64))
+        assert ours <= sp.fun * 1.02 + 1e-6, (ours, sp.fun)
