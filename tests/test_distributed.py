@@ -267,3 +267,48 @@ def test_demix_flat_obs_agent_split():
     assert float(img.reshape(-1)[-1]) == mod.NINF * mod.NINF - 1
     assert float(meta[0]) == mod.NINF * mod.NINF
     assert float(meta[-1]) == mod.OBS_DIM - 1
+
+
+def _la3_worker(rank, world, port, q):
+    from smartcal_amd.distributed.learner_actor import run_process
+    from smartcal_amd.envs.enet import ENetEnv
+    from smartcal_amd.rl.sac import Agent
+
+    torch.manual_seed(rank)
+    np.random.seed(rank)
+
+    def agent_factory():
+        return Agent(gamma=0.99, batch_size=4, n_actions=NACT, tau=0.005,
+                     max_mem_size=64, input_dims=[OBS_DIM], lr_a=1e-3,
+                     lr_c=1e-3, reward_scale=N, alpha=0.03,
+                     device=torch.device("cpu"))
+
+    def env_factory():
+        return ENetEnv(M, N, device=torch.device("cpu"))
+
+    scores = run_process(rank, world, agent_factory, env_factory,
+                         obs_dim=OBS_DIM, n_actions=NACT, episodes=2,
+                         epochs=1, steps=2, learner_addr="127.0.0.1",
+                         learner_port=port, max_transitions=6,
+                         backend="gloo")
+    if rank == 0:
+        q.put(scores)
+
+
+@pytest.mark.timeout(300)
+def test_learner_two_actors():
+    """1 learner + 2 actors (the reference's `mpirun -np 3` shape)."""
+    port = 29561
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_la3_worker, args=(r, 3, port, q))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+    assert all(p.exitcode == 0 for p in procs), \
+        [p.exitcode for p in procs]
+    scores = q.get()
+    assert len(scores) == 2
+    assert all(np.isfinite(s) for s in scores)
