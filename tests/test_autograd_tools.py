@@ -129,7 +129,7 @@ def test_inverse_hessian_vec_prod_direction():
     # 2/B X^T X ~ I so rho(I - H) << 1 and the normalized Neumann
     # iteration's fixed point matches H^{-1} v to first order
     X = torch.randn(200, n) * (0.5 ** 0.5)
-    yt = X @ torch.randn(n)
+    yt = (X @ torch.randn(n)).unsqueeze(1)   # match model output (B, 1)
     criterion = torch.nn.MSELoss()
     H = 2.0 / X.shape[0] * X.t() @ X
     v = torch.randn(n)

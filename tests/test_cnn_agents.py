@@ -165,7 +165,7 @@ def test_transformer_networks_trainable():
         loss = torch.nn.functional.mse_loss(crit(img, meta, act), target)
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < 0.3 * losses[0], (losses[0], losses[-1])
 
 
