@@ -70,7 +70,7 @@ def test_tsk_fit_simple():
         return (out - yt).norm() ** 2 / X.shape[0] \
             + 1e-4 * center_difference_loss(model) \
             + 1e-4 * sigma_loss(model)
-    l0 = float(lossfn())
+    l0 = float(lossfn().detach())
     for _ in range(150):
         opt.zero_grad()
         l = lossfn()
