@@ -76,7 +76,7 @@ def test_tsk_fit_simple():
         l = lossfn()
         l.backward()
         opt.step()
-    assert float(lossfn()) < 0.5 * l0
+    assert float(lossfn().detach()) < 0.5 * l0
 
 
 def test_trapmf():
