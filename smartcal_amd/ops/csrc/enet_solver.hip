@@ -125,7 +125,7 @@ __device__ static float ls_eval(const LdsLayout& L, const RegA& Ar,
 
 __device__ static float cubic_interp(float x1, float f1, float g1, float x2,
                                      float f2, float g2, float lo, float hi) {
-  // NaN/degenerate guards mirroring the reference line search\'s checks
+  // NaN/degenerate guards mirroring the reference line search's checks
   // (`lbfgsnew.py:556,624,673`): fall back to the bisection midpoint
   if (fabsf(x1 - x2) < 1e-20f || !isfinite(f1) || !isfinite(f2) ||
       !isfinite(g1) || !isfinite(g2))
@@ -497,7 +497,7 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
   if (lane < N) ys_[lane] = yg[(long)env * N + lane];
 
   // Q column cached per lane in REGISTERS: qc[m] = Q[m][lane] with
-  // Q = -2 A^T, i.e. qc[m] = -2 A[lane][m] (this lane\'s row of A).
+  // Q = -2 A^T, i.e. qc[m] = -2 A[lane][m] (this lane's row of A).
   // The two-loop recursion then runs on registers with one front-loaded
   // LDS broadcast per history entry instead of an lgkmcnt wait per FMA.
   __builtin_amdgcn_s_barrier();
