@@ -28,11 +28,18 @@ class VecENetEnv:
     """E independent elastic-net tuning problems, batch-stepped."""
 
     def __init__(self, E: int, M: int = 5, N: int = 15,
+                 reward_clamp: Optional[float] = None,
                  device: Optional[torch.device] = None):
+        """``reward_clamp``: optional symmetric clamp on rewards — the
+        reference leaves exactly this commented out with the note "only
+        useful for multiple environments" (``enetenv.py:150``); this IS
+        the multiple-environments case, so it is exposed here (off by
+        default for reward parity with the single env)."""
         self.E = int(E)
         self.M = int(M)
         self.N = int(N)
         self.K = 2
+        self.reward_clamp = reward_clamp
         self.device = device if device is not None else default_device()
         self.SNR = 0.1
         self.y: Optional[torch.Tensor] = None
@@ -86,6 +93,8 @@ class VecENetEnv:
         EE = torch.nan_to_num(EE, nan=0.0, posinf=1e6, neginf=-1e6)
         reward = torch.nan_to_num(reward, nan=-100.0, posinf=1e6,
                                   neginf=-1e6)
+        if self.reward_clamp is not None:
+            reward = reward.clamp(-self.reward_clamp, self.reward_clamp)
         obs = {"A": self.A.reshape(self.E, -1), "eig": EE}
         done = torch.zeros(self.E, dtype=torch.bool, device=self.device)
         return obs, reward, done, {}

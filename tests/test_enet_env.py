@@ -155,3 +155,8 @@ def test_vec_enet_env_contract():
     obs3, r3, *_ = env.step(np.array([[2.0, 2.0], [0.0, 0.0],
                                       [-2.0, -2.0]], dtype=np.float32))
     assert torch.isfinite(r3).all()
+    # optional clamp (the reference's commented-out multi-env clamp)
+    envc = VecENetEnv(2, 8, 8, reward_clamp=1.0)
+    envc.reset()
+    _, rc, *_ = envc.step(np.zeros((2, 2), np.float32))
+    assert (rc.abs() <= 1.0).all()
