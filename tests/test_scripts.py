@@ -38,6 +38,7 @@ def _run(script, *args, cwd):
     "scripts/demixing_rl/distributed_per_sac.py",
     "scripts/calibration/inspect_replaybuffer.py",
     "scripts/elasticnet/enet_eval.py",
+    "scripts/elasticnet/vec_sac.py",
 ])
 def test_script_help(script, tmp_path):
     r = _run(script, "--help", cwd=tmp_path)
@@ -145,3 +146,11 @@ def test_inspect_replaybuffer(tmp_path):
              "--out", "rb.png", cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
     assert (tmp_path / "rb.png").exists()
+
+
+def test_vec_sac_short(tmp_path):
+    """Vectorized-rollout SAC: E transitions per batched env step."""
+    r = _run("scripts/elasticnet/vec_sac.py", "--envs", "3",
+             "--iters", "8", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (tmp_path / "scores.pkl").exists()
