@@ -191,6 +191,28 @@ class ENetEnv(gymapi.Env):
         n_half = self.N // 2
         folds = [(slice(0, n_half), slice(n_half, self.N)),
                  (slice(n_half, self.N), slice(0, n_half))]
+
+        if self.device.type == "cuda":
+            # all 50 candidate fits (25-point grid x 2 folds) in ONE
+            # batched in-kernel solve — the serial closure path made a
+            # hint cost ~1000x more than an env step on a GPU
+            from smartcal_amd.ops import ext
+            pairs = [(l1, l2) for l1 in lam_grid for l2 in lam_grid]
+            Ab = torch.stack([A[tr] for _ in pairs for tr, _ in folds])
+            yb = torch.stack([y[tr] for _ in pairs for tr, _ in folds])
+            # kernel rho layout: (rho1=L2, rho2=L1), matching the loop
+            rho = torch.tensor([[l2, l1] for l1, l2 in pairs
+                                for _ in folds], device=A.device)
+            xs, _, _, _ = ext().enet_lbfgs_solve(
+                Ab.contiguous(), yb.contiguous(), rho.contiguous(), 5, 10, 7)
+            Ate = torch.stack([A[te] for _ in pairs for _, te in folds])
+            yte = torch.stack([y[te] for _ in pairs for _, te in folds])
+            r = torch.einsum("bnm,bm->bn", Ate, xs) - yte
+            mse = (r * r).mean(dim=1).reshape(len(pairs), 2).sum(dim=1)
+            best_pair = pairs[int(mse.argmin())]
+            hint_ = np.array(best_pair, dtype=np.float64)
+            return (hint_ - (HIGH + LOW) / 2) / ((HIGH - LOW) / 2)
+
         best = (None, float("inf"))
         for l1 in lam_grid:        # L1 weight (reference 'lambda1')
             for l2 in lam_grid:    # L2 weight (reference 'lambda2')

@@ -429,3 +429,41 @@ def test_transformer_agent_learn_gpu():
     torch.cuda.synchronize()
     a2 = np.asarray(agent.choose_action(s))
     assert np.isfinite(a2).all()
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason='needs GPU')
+def test_get_hint_batched_matches_cpu_quality():
+    """GPU get_hint (one batched 50-solve kernel launch) picks lambdas
+    whose CV error is as good as the serial CPU grid search's."""
+    from smartcal_amd.envs.enet import ENetEnv, LOW, HIGH
+    from smartcal_amd.ops import enet as enet_ops
+    torch.manual_seed(5)
+    env = ENetEnv(20, 20, provide_hint=True, device=torch.device("cuda"))
+    env.reset()
+    env._observe_y()
+    hint_gpu = env.get_hint()
+
+    env_cpu = ENetEnv.__new__(ENetEnv)
+    env_cpu.__dict__.update(env.__dict__)
+    env_cpu.device = torch.device("cpu")
+    env_cpu.A = env.A.cpu()
+    env_cpu.y = env.y.cpu()
+    env_cpu.y0 = env.y0.cpu()
+    hint_cpu = env_cpu.get_hint()
+
+    def cv_mse(hint):
+        lam = hint * (HIGH - LOW) / 2 + (HIGH + LOW) / 2
+        l1, l2 = float(lam[0]), float(lam[1])
+        A, y = env_cpu.A, env_cpu.y
+        n = 10
+        tot = 0.0
+        for tr, te in ((slice(0, n), slice(n, 20)),
+                       (slice(n, 20), slice(0, n))):
+            x, _ = enet_ops.lbfgs_solve_reference(A[tr], y[tr], rho1=l2,
+                                                  rho2=l1, epochs=5,
+                                                  max_iter=10)
+            r = A[te] @ x - y[te]
+            tot += float((r * r).mean())
+        return tot
+
+    assert cv_mse(hint_gpu) <= cv_mse(hint_cpu) * 1.05 + 1e-6
