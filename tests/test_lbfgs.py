@@ -178,3 +178,31 @@ def test_enet_solve_matches_scipy_lbfgsb():
             torch.from_numpy(A), torch.from_numpy(y), rho1, rho2)
         ours = f(x_t.numpy().astype(np.float64))
         assert ours <= sp.fun * 1.02 + 1e-6, (ours, sp.fun)
+
+
+def test_enet_solve_kkt_conditions():
+    """Approximate KKT optimality of the elastic-net solve: at the
+    solution, 2A^T(Ax-y) + 2 rho1 x + rho2 sign(x) ~ 0 on the support
+    and |2A^T(Ax-y)_i| <~ rho2 off it (subgradient condition)."""
+    import numpy as np
+    import torch
+    from smartcal_amd.ops import enet as enet_ops
+
+    rng = np.random.default_rng(1)
+    for trial in range(3):
+        N = M = 16
+        A = torch.from_numpy(rng.normal(size=(N, M)).astype(np.float32))
+        A = A / A.norm()
+        y = A @ torch.from_numpy(
+            (rng.normal(size=M) * (rng.random(M) > 0.5)).astype(np.float32))
+        rho1, rho2 = 0.03, 0.02
+        x, _ = enet_ops.lbfgs_solve_reference(A, y, rho1, rho2)
+        g_smooth = 2 * A.t() @ (A @ x - y) + 2 * rho1 * x
+        scale = float(g_smooth.abs().max().clamp(min=1.0))
+        on = x.abs() > 1e-3
+        if on.any():
+            kkt_on = (g_smooth[on] + rho2 * torch.sign(x[on])).abs().max()
+            assert float(kkt_on) < 0.05 * scale + 0.02, float(kkt_on)
+        if (~on).any():
+            kkt_off = g_smooth[~on].abs().max()
+            assert float(kkt_off) <= rho2 * 1.5 + 0.02, float(kkt_off)
