@@ -467,3 +467,32 @@ def test_get_hint_batched_matches_cpu_quality():
         return tot
 
     assert cv_mse(hint_gpu) <= cv_mse(hint_cpu) * 1.05 + 1e-6
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason='needs GPU')
+def test_enet_kernel_kkt_conditions():
+    """Approximate KKT optimality of the in-kernel solve across the
+    rho range (incl. the small-rho regime the guards target)."""
+    from smartcal_amd import ops
+    torch.manual_seed(9)
+    for rho1, rho2 in ((0.05, 0.02), (0.005, 0.002), (0.001, 0.001)):
+        N = M = 20
+        A = torch.randn(N, M, device="cuda")
+        A = A / A.norm()
+        y = A @ (torch.randn(M, device="cuda")
+                 * (torch.rand(M, device="cuda") > 0.5))
+        rho = torch.tensor([[rho1, rho2]], device="cuda")
+        xg, *_ = ops.ext().enet_lbfgs_solve(
+            A.unsqueeze(0).contiguous(), y.unsqueeze(0).contiguous(),
+            rho, 20, 10, 7)
+        x = xg[0]
+        g_smooth = 2 * A.t() @ (A @ x - y) + 2 * rho1 * x
+        scale = float(g_smooth.abs().max().clamp(min=1.0))
+        on = x.abs() > 1e-3
+        if on.any():
+            kkt_on = (g_smooth[on] + rho2 * torch.sign(x[on])).abs().max()
+            assert float(kkt_on) < 0.08 * scale + 0.03, (rho1, float(kkt_on))
+        if (~on).any():
+            kkt_off = g_smooth[~on].abs().max()
+            assert float(kkt_off) <= rho2 * 2.0 + 0.03, (rho1,
+                                                         float(kkt_off))
