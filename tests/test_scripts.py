@@ -39,6 +39,16 @@ def _run(script, *args, cwd):
     "scripts/calibration/inspect_replaybuffer.py",
     "scripts/elasticnet/enet_eval.py",
     "scripts/elasticnet/vec_sac.py",
+    "scripts/calibration/pipeline.py",
+    "scripts/calibration/analysis.py",
+    "scripts/demixing_rl/evaluate_tsk_msp.py",
+    "scripts/demixing_rl/influence_tsk.py",
+    "scripts/demixing_rl/plot_tsk.py",
+    "scripts/demixing_rl/plot_databuffer.py",
+    "scripts/demixing/eval_model.py",
+    "scripts/demixing/evaluate.py",
+    "scripts/demixing/mergebuffers.py",
+    "scripts/demixing/populatebuffer.py",
 ])
 def test_script_help(script, tmp_path):
     r = _run(script, "--help", cwd=tmp_path)
