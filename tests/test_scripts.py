@@ -164,3 +164,14 @@ def test_vec_sac_short(tmp_path):
              "--iters", "8", cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
     assert (tmp_path / "scores.pkl").exists()
+
+
+def test_calibration_pipeline_tiny(tmp_path):
+    """doall.sh-equivalent end to end at toy size: simulate -> calibrate
+    -> influence -> mean image, files written."""
+    r = _run("scripts/calibration/pipeline.py", "--K", "2", "--stations",
+             "8", "--nf", "2", "--ts", "1", "--tdelta", "4", "--admm", "2",
+             "--poly", "2", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    made = list(tmp_path.iterdir())
+    assert made, "pipeline wrote no outputs"
