@@ -82,9 +82,15 @@ def test_demix_fuzzy_short(tmp_path):
 
 
 def test_supervised_pipeline(tmp_path):
-    """simulate_data → train_model end to end at toy size."""
+    """simulate_data → (merge/rebalance) → train_model at toy size."""
     r = _run("scripts/demixing/simulate_data.py", "--samples", "2",
              "--ninf", "16", "--stations", "6", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = _run("scripts/demixing/mergebuffers.py", "simul_data.buffer",
+             "simul_data.buffer", "--out", "merged.buffer", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = _run("scripts/demixing/populatebuffer.py", "simul_data.buffer",
+             "--out", "balanced.buffer", cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
     r = _run("scripts/demixing/train_model.py", "--iters", "3",
              "--batch", "2", "--ninf", "16", cwd=tmp_path)
@@ -102,6 +108,13 @@ def test_distill_pipeline(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     r = _run("scripts/demixing_rl/train_tsk.py", "--iters", "5",
              "--batch", "2", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    # downstream tools consume the artifacts just written
+    r = _run("scripts/demixing_rl/plot_tsk.py", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = _run("scripts/demixing_rl/plot_databuffer.py", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = _run("scripts/demixing_rl/influence_tsk.py", cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
 
 
