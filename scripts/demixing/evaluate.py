@@ -24,13 +24,15 @@ def main():
     ap.add_argument("--model", default="transformer.model")
     ap.add_argument("--ninf", default=64, type=int)
     ap.add_argument("--seed", default=0, type=int)
+    ap.add_argument("--stations", default=62, type=int)
     args = ap.parse_args()
     device = default_device()
     rng = np.random.default_rng(args.seed)
 
     K = 6
     Nout = args.ninf * args.ninf + 8
-    x, y, _ = generate_training_example(rng, Ninf=args.ninf)
+    x, y, _ = generate_training_example(rng, Ninf=args.ninf,
+                                        N_stations=args.stations)
     net = TransformerEncoder(num_layers=1, input_dim=K * Nout,
                              model_dim=K * (args.ninf + 2),
                              num_classes=K - 1, num_heads=K).to(device)

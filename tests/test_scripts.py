@@ -96,6 +96,12 @@ def test_supervised_pipeline(tmp_path):
              "--batch", "2", "--ninf", "16", cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
     assert (tmp_path / "transformer.model").exists()
+    r = _run("scripts/demixing/eval_model.py", "--ninf", "16",
+             "--samples", "2", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = _run("scripts/demixing/evaluate.py", "--ninf", "16",
+             "--stations", "6", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
 
 
 def test_distill_pipeline(tmp_path):
