@@ -122,6 +122,9 @@ def test_distill_pipeline(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     r = _run("scripts/demixing_rl/influence_tsk.py", cwd=tmp_path)
     assert r.returncode == 0, r.stderr[-2000:]
+    r = _run("scripts/demixing_rl/evaluate_tsk_msp.py", "--episodes", "1",
+             "--stations", "6", cwd=tmp_path)
+    assert r.returncode == 0, r.stderr[-2000:]
 
 
 def test_analysis_cli(tmp_path):
