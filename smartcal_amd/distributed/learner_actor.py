@@ -196,7 +196,12 @@ def run_process(rank: int, world_size: int, agent_factory, env_factory,
     """Entry point preserving the reference CLI semantics
     (`distributed_per_sac.py:154-194`)."""
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # Composite backend: the flat weight broadcast rides RCCL over
+        # xGMI when the agent lives on a GPU, while the hdr/transition
+        # gathers are CPU tensors and need gloo — a NCCL-only group
+        # cannot carry them.
+        backend = "cpu:gloo,cuda:nccl" if torch.cuda.is_available() \
+            else "gloo"
     if not dist.is_initialized():
         dist.init_process_group(
             backend=backend, rank=rank, world_size=world_size,

@@ -75,8 +75,12 @@ class VecENetEnv:
     def step(self, actions, keepnoise: bool = False):
         """actions (E, 2) in [-1, 1] → (obs dict, rewards (E,), done
         (E,) bool, info). Same per-instance semantics as ENetEnv.step."""
-        a = torch.as_tensor(np.asarray(actions), dtype=torch.float32) \
-            .reshape(self.E, self.K)
+        if torch.is_tensor(actions):
+            a = actions.detach().to(dtype=torch.float32) \
+                .reshape(self.E, self.K)
+        else:
+            a = torch.as_tensor(np.asarray(actions), dtype=torch.float32) \
+                .reshape(self.E, self.K)
         if a.device != self.device:
             a = a.to(self.device)
         a = torch.nan_to_num(a.detach(), nan=0.0, posinf=1.0, neginf=-1.0)

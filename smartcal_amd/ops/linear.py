@@ -14,6 +14,17 @@ dgamma/dbeta column reductions, (2) dX = dZ @ W, (3) dW = dZ^T @ X + db —
 both MFMA f32 GEMMs.
 
 CPU tensors run the equivalent torch composition (also the test oracle).
+
+CONTRACT RESTRICTION (deliberate): on the HIP path the backward
+accumulates parameter gradients straight into ``.grad`` (the agents' flat
+gradient pools) and returns ``None`` for the W/b/gamma/beta grad outputs.
+This removes every per-parameter AccumulateGrad/zero kernel from the
+learn graph, but it means ``torch.autograd.grad(loss, params)`` and
+double-backward (``create_graph=True`` — e.g.
+``autograd_tools.hessian_vec_prod``) are NOT supported through the HIP
+fused path; influence/HVP tooling uses plain ``nn.Linear`` models, which
+is what every in-repo consumer of ``autograd_tools`` does. The CPU path
+has no such restriction.
 """
 
 from __future__ import annotations

@@ -108,14 +108,27 @@ class ReplayBuffer:
         with open(fn, "rb") as f:
             sd = pickle.load(f)
         for k, v in sd.items():
+            if k == "mem_size":
+                # capacity is a property of THIS buffer's tensors, not of
+                # the checkpoint: restoring a larger saved mem_size would
+                # desync ring indexing from the allocated rows
+                continue
             cur = getattr(self, k, None)
             if torch.is_tensor(v) and torch.is_tensor(cur):
                 if v.shape == cur.shape:
                     cur.copy_(v.to(self.device))
-                else:       # filled-prefix checkpoint
-                    cur[:v.shape[0]].copy_(v.to(self.device))
+                else:       # filled-prefix checkpoint; a prefix longer
+                    # than the current capacity (checkpoint from a larger
+                    # buffer) keeps only what fits
+                    n = min(v.shape[0], cur.shape[0])
+                    cur[:n].copy_(v[:n].to(self.device))
             else:
                 setattr(self, k, v)
+        # a counter from a larger buffer over-claims filled entries the
+        # copy above could not keep — clamp so len() stays honest
+        if sd.get("mem_size", self.mem_size) > self.mem_size \
+                and self.mem_cntr > self.mem_size:
+            self.mem_cntr = self.mem_size
 
 
 class SumTree:

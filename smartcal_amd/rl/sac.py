@@ -167,6 +167,25 @@ class Agent:
     def _pair(self, f1, f2):
         return self._fork(f1, f2)
 
+    def _sync_side_streams(self):
+        """Join every _fork side stream into the current stream. Backward
+        of a forward that ran on a side stream also runs there
+        (stream-aware autograd), and the fused-linear backward writes
+        ``flat_grad`` directly — bypassing AccumulateGrad, so the
+        engine's leaf-stream sync does not cover it. Without this join a
+        main-stream grad all-reduce (data-parallel) could read
+        partially-written gradients. Event record/wait is graph-capture
+        safe (captured as dependency edges)."""
+        if self.device.type != "cuda" or not hasattr(self, "_side_streams"):
+            return
+        if not hasattr(self, "_sync_evs"):
+            self._sync_evs = []
+        while len(self._sync_evs) < len(self._side_streams):
+            self._sync_evs.append(torch.cuda.Event())
+        for ev, s in zip(self._sync_evs, self._side_streams):
+            ev.record(s)
+            ev.wait()
+
     def _learn_body(self, state_batch, new_state_batch, action_batch,
                     reward_batch, terminal_batch, hint_batch, is_w=None):
         """Tensor-only learn step (hipGraph-capturable on GPU): soft target,
@@ -206,6 +225,7 @@ class Agent:
         self.critic_1_opt.zero_grad()
         self.critic_2_opt.zero_grad()
         critic_loss.backward()
+        self._sync_side_streams()
         self._grad_sync([self.critic_1_fp, self.critic_2_fp])
         self._fork(self.critic_1_opt.step, self.critic_2_opt.step)
 
@@ -238,6 +258,7 @@ class Agent:
             p.requires_grad_(True)
         for p in self.critic_2.parameters():
             p.requires_grad_(True)
+        self._sync_side_streams()
         self._grad_sync([self.actor_fp])
         self.actor_opt.step()
         self.update_network_parameters()
