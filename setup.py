@@ -26,6 +26,7 @@ sources = [str(CSRC / f) for f in [
     "enet_solver.hip",
     "als_sweep.hip",
     "conv2d.hip",
+    "per.hip",
 ]]
 
 setup(

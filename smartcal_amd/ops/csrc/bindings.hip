@@ -54,6 +54,12 @@ extern "C" __global__ void als_sweep_kernel(
 extern "C" __global__ void enet_influence_kernel(
     const float*, const float*, const float*, const float*, const float*,
     const int*, const float*, float*, float*, int, int, int);
+extern "C" __global__ void per_sample_kernel(const float*, const float*,
+                                             long*, float*, float*, int, int,
+                                             float);
+extern "C" __global__ void per_update_kernel(float*, const long*,
+                                             const float*, int, float, float,
+                                             float);
 
 namespace {
 
@@ -434,6 +440,47 @@ std::tuple<at::Tensor, at::Tensor> enet_influence(
   return {EE, reward};
 }
 
+// PER sampling capacity: n fp32 prefix sums must fit in LDS next to the
+// scan scratch and weight buffer. 32k priorities = 128 KB of the CU's
+// 160 KB — covers every workload config (mem_size <= 16000); larger
+// buffers take the torch scan path in python (ops/per.py).
+#define PER_LDS_MAX_N 32768
+#define PER_MAX_B 4096
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> per_sample(
+    const at::Tensor& priorities, const at::Tensor& u, double beta) {
+  check_f32(priorities, "priorities");
+  check_f32(u, "u");
+  const int n = priorities.numel();
+  const int B = u.numel();
+  TORCH_CHECK(n >= 1 && n <= PER_LDS_MAX_N,
+              "per_sample: n must be in [1, 32768]");
+  TORCH_CHECK(B >= 1 && B <= PER_MAX_B, "per_sample: B must be in [1, 4096]");
+  auto idx = at::empty({B}, priorities.options().dtype(at::kLong));
+  auto probs = at::empty({B}, priorities.options());
+  auto w = at::empty({B}, priorities.options());
+  const int T = 256;
+  const size_t lds = (size_t)(n + T + 1 + B) * sizeof(float);
+  hipLaunchKernelGGL(per_sample_kernel, dim3(1), dim3(T), lds, stream(),
+                     priorities.data_ptr<float>(), u.data_ptr<float>(),
+                     idx.data_ptr<long>(), probs.data_ptr<float>(),
+                     w.data_ptr<float>(), n, B, (float)beta);
+  return {idx, probs, w};
+}
+
+void per_update(at::Tensor& priorities, const at::Tensor& idx,
+                const at::Tensor& td, double eps, double alpha,
+                double max_priority) {
+  check_f32(priorities, "priorities");
+  check_f32(td, "td");
+  TORCH_CHECK(idx.scalar_type() == at::kLong, "idx must be int64");
+  const int B = idx.numel();
+  hipLaunchKernelGGL(per_update_kernel, dim3((B + 255) / 256), dim3(256), 0,
+                     stream(), priorities.data_ptr<float>(),
+                     idx.data_ptr<long>(), td.data_ptr<float>(), B,
+                     (float)eps, (float)alpha, (float)max_priority);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -452,4 +499,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam);
   m.def("enet_lbfgs_solve", &enet_lbfgs_solve);
   m.def("enet_influence", &enet_influence);
+  m.def("per_sample", &per_sample);
+  m.def("per_update", &per_update);
 }

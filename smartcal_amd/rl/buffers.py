@@ -206,15 +206,14 @@ class PERBuffer(ReplayBuffer):
     def sample_buffer(self, batch_size: int):
         n = len(self)
         pri = self.priorities[:n]
-        idx, probs = per_ops.stratified_sample(pri, batch_size)
         self.beta = min(1.0, self.beta + self.BETA_INC)
-        weights = per_ops.importance_weights(probs, n, self.beta)
+        idx, probs, weights = per_ops.sample_with_weights(pri, batch_size,
+                                                          self.beta)
         batch = (self.state_memory[idx], self.action_memory[idx],
                  self.reward_memory[idx], self.new_state_memory[idx],
                  self.terminal_memory[idx], self.hint_memory[idx])
         return batch, idx, weights
 
     def update_priorities(self, idx: torch.Tensor, td_errors: torch.Tensor):
-        pri = (td_errors.detach().abs().reshape(-1) + self.EPS) \
-            .clamp(max=self.MAX_PRIORITY).pow(self.ALPHA)
-        self.priorities[idx] = pri.to(self.priorities.dtype)
+        per_ops.update_priorities(self.priorities, idx, td_errors, self.EPS,
+                                  self.ALPHA, self.MAX_PRIORITY)

@@ -145,10 +145,9 @@ class DictPERBuffer(DictReplayBuffer):
 
     def sample_buffer(self, batch_size: int):
         n = len(self)
-        idx, probs = per_ops.stratified_sample(self.priorities[:n],
-                                               batch_size)
         self.beta = min(1.0, self.beta + self.BETA_INC)
-        weights = per_ops.importance_weights(probs, n, self.beta)
+        idx, probs, weights = per_ops.sample_with_weights(
+            self.priorities[:n], batch_size, self.beta)
         batch = self._gather(idx)
         if self.normalize_reward:
             # reference `demix_td3.py:162-166`: standardize sampled rewards
@@ -159,6 +158,5 @@ class DictPERBuffer(DictReplayBuffer):
         return batch, idx, weights
 
     def update_priorities(self, idx, td_errors):
-        pri = (td_errors.detach().abs().reshape(-1) + self.EPS) \
-            .clamp(max=self.MAX_PRIORITY).pow(self.ALPHA)
-        self.priorities[idx] = pri.to(self.priorities.dtype)
+        per_ops.update_priorities(self.priorities, idx, td_errors, self.EPS,
+                                  self.ALPHA, self.MAX_PRIORITY)

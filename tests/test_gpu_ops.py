@@ -499,3 +499,58 @@ def test_enet_kernel_kkt_conditions():
             kkt_off = g_smooth[~on].abs().max()
             assert float(kkt_off) <= rho2 * 2.0 + 0.03, (rho1,
                                                          float(kkt_off))
+
+
+def test_per_sample_kernel_matches_torch_oracle():
+    """per_sample_kernel vs the torch cumsum+searchsorted oracle.
+
+    Integer-valued priorities keep every partial sum exact in fp32 (total
+    << 2^24), so the stratified indices must match EXACTLY; probs and
+    max-normalized IS weights to fp32 tolerance."""
+    from smartcal_amd.ops import per as per_ops
+    for n, B in [(7, 4), (64, 64), (1024, 64), (16000, 256), (1, 8)]:
+        torch.manual_seed(n + B)
+        pri = torch.randint(1, 11, (n,), device=DEV).float()
+        pri[torch.rand(n, device=DEV) < 0.2] = 0.0  # zero-priority runs
+        if float(pri.sum()) == 0.0:
+            pri[0] = 1.0
+        u = torch.rand(B, device=DEV)
+        beta = 0.62
+        idx_g, probs_g, w_g = ops.ext().per_sample(pri, u, beta)
+        idx_t, probs_t = per_ops._torch_stratified(pri, B, u)
+        w_t = per_ops.importance_weights(probs_t, n, beta)
+        assert torch.equal(idx_g, idx_t), (n, B)
+        assert torch.allclose(probs_g, probs_t, atol=1e-7)
+        assert torch.allclose(w_g, w_t, atol=1e-5, rtol=1e-5)
+
+
+def test_per_update_kernel_matches_torch():
+    from smartcal_amd.ops import per as per_ops
+    torch.manual_seed(3)
+    n, B = 512, 64
+    pri = torch.rand(n, device=DEV) + 0.1
+    pri_ref = pri.clone()
+    idx = torch.randperm(n, device=DEV)[:B]
+    td = torch.randn(B, device=DEV) * 2
+    ops.ext().per_update(pri, idx, td.contiguous(), 0.01, 0.6, 1.0)
+    ref = (td.abs() + 0.01).clamp(max=1.0).pow(0.6)
+    pri_ref[idx] = ref
+    assert torch.allclose(pri, pri_ref, atol=1e-6)
+
+
+def test_per_buffer_gpu_uses_kernel_end_to_end():
+    """PERBuffer on GPU: sample indices within range, weights in (0,1],
+    priorities updated on device, distribution sane."""
+    from smartcal_amd.rl.buffers import PERBuffer
+    torch.manual_seed(5)
+    buf = PERBuffer(256, [12], 2, device=DEV)
+    for i in range(300):
+        s = torch.randn(12)
+        buf.store_transition(s, torch.randn(2), float(i), s, False,
+                             torch.zeros(2))
+    (state, act, rew, state_, done, hint), idx, w = buf.sample_buffer(64)
+    assert state.shape == (64, 12) and idx.shape == (64,)
+    assert idx.min() >= 0 and idx.max() < 256
+    assert float(w.max()) <= 1.0 + 1e-6 and float(w.min()) > 0
+    buf.update_priorities(idx, torch.randn(64, 1, device=DEV))
+    assert torch.isfinite(buf.priorities).all()
