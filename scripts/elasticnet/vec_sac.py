@@ -52,10 +52,9 @@ def main():
                                                    reparameterize=False)
         obs_, rewards, done, _ = env.step(actions.to(env.device))
         states_ = torch.cat((obs_["eig"], obs_["A"]), dim=1)
-        for e in range(args.envs):
-            agent.replaymem.store_transition(
-                states[e].cpu(), actions[e].cpu(), float(rewards[e]),
-                states_[e].cpu(), False, zeros2)
+        # one batched device-resident write for all E transitions
+        agent.replaymem.store_batch(states, actions,
+                                    torch.as_tensor(rewards), states_, done)
         agent.learn()
         scores.append(float(rewards.mean()))
         if (it + 1) % args.reset_every == 0:

@@ -77,6 +77,33 @@ class ReplayBuffer:
             self.hint_memory[i] = self._as(hint, self.hint_memory[i])
         self.mem_cntr += 1
 
+    def store_batch(self, states, actions, rewards, states_, dones,
+                    hints=None):
+        """Vectorized store of E transitions in one shot (device-resident,
+        no per-transition host sync) — the VecENetEnv rollout path. Ring
+        semantics identical to E sequential store_transition calls."""
+        E = states.shape[0]
+        idx = (torch.arange(E, device=self.device)
+               + self.mem_cntr) % self.mem_size
+        self.state_memory[idx] = states.detach().to(
+            self.device, torch.float32).reshape(E, -1)
+        self.new_state_memory[idx] = states_.detach().to(
+            self.device, torch.float32).reshape(E, -1)
+        self.action_memory[idx] = actions.detach().to(
+            self.device, torch.float32).reshape(E, -1)
+        self.reward_memory[idx] = rewards.detach().to(
+            self.device, torch.float32).reshape(-1)
+        if torch.is_tensor(dones):
+            self.terminal_memory[idx] = dones.to(self.device,
+                                                 torch.bool).reshape(-1)
+        else:
+            self.terminal_memory[idx] = bool(dones)
+        if hints is not None:
+            self.hint_memory[idx] = hints.detach().to(
+                self.device, torch.float32).reshape(E, -1)
+        self.mem_cntr += E
+        return idx
+
     def sample_buffer(self, batch_size: int):
         max_mem = len(self)
         idx = torch.randint(0, max_mem, (batch_size,), device=self.device)
@@ -202,6 +229,17 @@ class PERBuffer(ReplayBuffer):
         super().store_transition(state, action, reward, state_, done, hint)
         mx = float(self.priorities.max()) if len(self) > 1 else 0.0
         self.priorities[i] = mx if mx > 0 else self.MAX_PRIORITY
+
+    def store_batch(self, states, actions, rewards, states_, dones,
+                    hints=None):
+        had = len(self) > 1
+        idx = super().store_batch(states, actions, rewards, states_, dones,
+                                  hints)
+        mx = self.priorities.max().clamp(min=0.0) if had \
+            else torch.tensor(0.0, device=self.device)
+        self.priorities[idx] = torch.where(
+            mx > 0, mx, torch.tensor(self.MAX_PRIORITY, device=self.device))
+        return idx
 
     def sample_buffer(self, batch_size: int):
         n = len(self)
