@@ -10,7 +10,11 @@ Writes gpurun_out/vec_sweep.json.
 
 import json
 import os
+import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
 
 import torch
 
