@@ -47,7 +47,14 @@ def main():
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=32)
     ap.add_argument("--seed", type=int, default=1)
+    # BASELINE config 2: elastic-net TD3 + prioritized replay, bf16
+    ap.add_argument("--algo", choices=["sac", "td3"], default="sac")
+    ap.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32")
     args = ap.parse_args()
+
+    if args.dtype == "bf16":
+        from smartcal_amd.ops import linear as linear_ops
+        linear_ops.set_compute_dtype("bf16")
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -74,10 +81,20 @@ def main():
                 dist.all_reduce(fp.flat_grad, op=dist.ReduceOp.SUM)
                 fp.flat_grad.div_(world)
 
-    agent = Agent(gamma=0.99, batch_size=64, n_actions=2, tau=0.005,
-                  max_mem_size=1024, input_dims=[N + N * M], lr_a=1e-3,
-                  lr_c=1e-3, reward_scale=N, alpha=0.03, prioritized=False,
-                  use_hint=False, device=device, grad_hook=grad_hook)
+    if args.algo == "td3":
+        # reference elasticnet/main_td3.py:12-22 config + PER
+        from smartcal_amd.rl.td3 import Agent as TD3Agent
+        agent = TD3Agent(gamma=0.99, batch_size=64, n_actions=2, tau=0.005,
+                         max_mem_size=1024, input_dims=[N + N * M],
+                         lr_a=1e-3, lr_c=1e-3, warmup=0, noise=0.1,
+                         prioritized=True, use_hint=False, device=device,
+                         grad_hook=grad_hook)
+    else:
+        agent = Agent(gamma=0.99, batch_size=64, n_actions=2, tau=0.005,
+                      max_mem_size=1024, input_dims=[N + N * M], lr_a=1e-3,
+                      lr_c=1e-3, reward_scale=N, alpha=0.03,
+                      prioritized=False, use_hint=False, device=device,
+                      grad_hook=grad_hook)
 
     obs = env.reset()
     zeros2 = np.zeros(2, dtype=np.float32)
@@ -165,8 +182,9 @@ def main():
     ms_per_step = elapsed_max / args.steps * 1000.0
 
     if rank == 0:
+        algo = args.algo.upper()
         print(json.dumps({
-            "metric": "env-steps/sec (whole node), elastic-net SAC",
+            "metric": f"env-steps/sec (whole node), elastic-net {algo}",
             "value": value,
             "unit": "env-steps/s",
             "n_gpus": n_gpus,
@@ -178,11 +196,12 @@ def main():
             "vs_baseline": None,
             "band_steps_per_s": [round(min(chunk_rates) * max(world, 1), 1),
                                  round(max(chunk_rates) * max(world, 1), 1)],
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
-            "config": {"model": "elasticnet-sac", "N": N, "M": M,
+            "config": {"model": f"elasticnet-{args.algo}", "N": N, "M": M,
                        "input_dims": N + N * M, "global_batch": 64 * max(world, 1),
                        "replay": 1024, "reward_scale": N, "alpha": 0.03,
+                       "prioritized": args.algo == "td3",
                        "parallelism": f"dp{max(world, 1)}"},
         }))
 

@@ -27,6 +27,7 @@ sources = [str(CSRC / f) for f in [
     "als_sweep.hip",
     "conv2d.hip",
     "per.hip",
+    "fused_linear_bf16.hip",
 ]]
 
 setup(

@@ -57,6 +57,13 @@ extern "C" __global__ void enet_influence_kernel(
 extern "C" __global__ void per_sample_kernel(const float*, const float*,
                                              long*, float*, float*, int, int,
                                              float);
+extern "C" __global__ void fused_linear_bf16_fwd_kernel(
+    const float*, const float*, const float*, const float*, const float*,
+    float*, float*, float*, int, int, int, int, int);
+extern "C" __global__ void gemm_bf16_nn_kernel(const float*, const float*,
+                                               float*, int, int, int);
+extern "C" __global__ void gemm_bf16_tn_kernel(const float*, const float*,
+                                               float*, int, int, int, int);
 extern "C" __global__ void per_update_kernel(float*, const long*,
                                              const float*, int, float, float,
                                              float);
@@ -346,6 +353,67 @@ void mfma_gemm_tn_bias_into(const at::Tensor& dz, const at::Tensor& x,
                      Bb, N, 1);
 }
 
+// ---- bf16-compute variants (fp32 interfaces, bf16 MFMA inside) ----
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> fused_linear_bf16_fwd(
+    const at::Tensor& x, const at::Tensor& W,
+    const c10::optional<at::Tensor>& bias,
+    const c10::optional<at::Tensor>& gamma,
+    const c10::optional<at::Tensor>& beta, int64_t act, bool with_ln) {
+  check_f32(x, "x");
+  check_f32(W, "W");
+  const int B = x.size(0), K = x.size(1), N = W.size(0);
+  TORCH_CHECK(W.size(1) == K, "W/K mismatch");
+  TORCH_CHECK(N <= 1536, "fused_linear_bf16: N>1536 unsupported");
+  TORCH_CHECK(!with_ln || gamma.has_value(), "LN requires gamma");
+  auto y = at::empty({B, N}, x.options());
+  auto zhat = with_ln ? at::empty({B, N}, x.options())
+                      : at::empty({0}, x.options());
+  auto rstd = with_ln ? at::empty({B}, x.options())
+                      : at::empty({0}, x.options());
+  dim3 grid((B + 15) / 16);
+  hipLaunchKernelGGL(fused_linear_bf16_fwd_kernel, grid, dim3(512), 0,
+                     stream(), x.data_ptr<float>(), W.data_ptr<float>(),
+                     bias ? bias->data_ptr<float>() : nullptr,
+                     gamma ? gamma->data_ptr<float>() : nullptr,
+                     beta ? beta->data_ptr<float>() : nullptr,
+                     y.data_ptr<float>(),
+                     with_ln ? zhat.data_ptr<float>() : nullptr,
+                     with_ln ? rstd.data_ptr<float>() : nullptr, B, K, N,
+                     (int)act, with_ln ? 1 : 0);
+  return {y, zhat, rstd};
+}
+
+at::Tensor mfma_gemm_nn_bf16(const at::Tensor& A, const at::Tensor& B) {
+  check_f32(A, "A");
+  check_f32(B, "B");
+  const int M = A.size(0), K = A.size(1), N = B.size(1);
+  TORCH_CHECK(B.size(0) == K, "gemm_nn shape mismatch");
+  auto C = at::empty({M, N}, A.options());
+  dim3 grid((M + 15) / 16, (N + 63) / 64);
+  hipLaunchKernelGGL(gemm_bf16_nn_kernel, grid, dim3(256), 0, stream(),
+                     A.data_ptr<float>(), B.data_ptr<float>(),
+                     C.data_ptr<float>(), M, K, N);
+  return C;
+}
+
+void mfma_gemm_tn_bias_into_bf16(const at::Tensor& dz, const at::Tensor& x,
+                                 at::Tensor dW, at::Tensor db) {
+  check_f32(dz, "dz");
+  check_f32(x, "x");
+  const int Bb = dz.size(0), N = dz.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == Bb, "gemm_tn batch mismatch");
+  TORCH_CHECK(dW.is_contiguous() && db.is_contiguous(),
+              "grad views must be contiguous");
+  dim3 grid((N + 15) / 16, (K + 63) / 64);
+  hipLaunchKernelGGL(gemm_bf16_tn_kernel, grid, dim3(256), 0, stream(),
+                     dz.data_ptr<float>(), x.data_ptr<float>(),
+                     dW.data_ptr<float>(), N, Bb, K, 1);
+  hipLaunchKernelGGL(colsum_kernel, dim3((N + 63) / 64), dim3(256), 0,
+                     stream(), dz.data_ptr<float>(), db.data_ptr<float>(),
+                     Bb, N, 1);
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor> tanh_gauss_fwd(
     const at::Tensor& mu, const at::Tensor& logsigma, const at::Tensor& eps,
     double max_action) {
@@ -501,4 +569,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("enet_influence", &enet_influence);
   m.def("per_sample", &per_sample);
   m.def("per_update", &per_update);
+  m.def("fused_linear_bf16_fwd", &fused_linear_bf16_fwd);
+  m.def("mfma_gemm_nn_bf16", &mfma_gemm_nn_bf16);
+  m.def("mfma_gemm_tn_bias_into_bf16", &mfma_gemm_tn_bias_into_bf16);
 }

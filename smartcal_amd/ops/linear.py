@@ -44,6 +44,22 @@ ACT_TANH = 3
 _ACT_CODES = {"none": ACT_NONE, "elu": ACT_ELU, "relu": ACT_RELU,
               "tanh": ACT_TANH}
 
+# Process-wide compute dtype for the fused-linear family ("fp32" | "bf16").
+# bf16 keeps fp32 master weights, fp32 tensors and fp32 LayerNorm/epilogue;
+# only the GEMM multiplies run on the bf16 MFMA pipe
+# (v_mfma_f32_16x16x32_bf16, fp32 accumulate) — BASELINE config 2.
+_COMPUTE_DTYPE = "fp32"
+
+
+def set_compute_dtype(dtype: str) -> None:
+    global _COMPUTE_DTYPE
+    assert dtype in ("fp32", "bf16"), dtype
+    _COMPUTE_DTYPE = dtype
+
+
+def get_compute_dtype() -> str:
+    return _COMPUTE_DTYPE
+
 
 _scratch: dict = {}
 
@@ -77,14 +93,17 @@ def _grad_view(p: torch.Tensor) -> torch.Tensor:
 class _FusedLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, W, b, gamma, beta, act_code: int, with_ln: bool):
-        y, zhat, rstd = ext().fused_linear_fwd(x, W, b, gamma, beta,
-                                               act_code, with_ln)
+        bf16 = _COMPUTE_DTYPE == "bf16"
+        fwd = ext().fused_linear_bf16_fwd if bf16 \
+            else ext().fused_linear_fwd
+        y, zhat, rstd = fwd(x, W, b, gamma, beta, act_code, with_ln)
         ctx.save_for_backward(x, W, gamma, zhat, rstd, y)
         # python refs to the parameter objects for direct grad
         # accumulation (not part of the autograd graph)
         ctx.param_refs = (W, b, gamma, beta)
         ctx.act_code = act_code
         ctx.with_ln = with_ln
+        ctx.bf16 = bf16
         return y
 
     @staticmethod
@@ -108,12 +127,18 @@ class _FusedLinearFn(torch.autograd.Function):
         else:
             dz, _, _ = ext().fused_linear_bwd_dz(
                 dy, y, zhat, rstd, gamma, ctx.act_code, False)
-        dx = ext().mfma_gemm_nn(dz, W)          # (B,N) @ (N,K) -> (B,K)
-        if want_w:
-            # dW = dz^T @ x and db = col-sum(dz), accumulated straight
-            # into the flat gradient pool by the kernels
-            ext().mfma_gemm_tn_bias_into(dz, x, _grad_view(Wp),
-                                         _grad_view(bp))
+        if getattr(ctx, "bf16", False):
+            dx = ext().mfma_gemm_nn_bf16(dz, W)
+            if want_w:
+                ext().mfma_gemm_tn_bias_into_bf16(dz, x, _grad_view(Wp),
+                                                  _grad_view(bp))
+        else:
+            dx = ext().mfma_gemm_nn(dz, W)      # (B,N) @ (N,K) -> (B,K)
+            if want_w:
+                # dW = dz^T @ x and db = col-sum(dz), accumulated straight
+                # into the flat gradient pool by the kernels
+                ext().mfma_gemm_tn_bias_into(dz, x, _grad_view(Wp),
+                                             _grad_view(bp))
         return dx, None, None, None, None, None, None
 
 
@@ -241,7 +266,7 @@ def fused_chain(x: torch.Tensor, layers) -> torch.Tensor:
     squeeze = x.dim() == 1
     if squeeze:
         x = x.unsqueeze(0)
-    if use_hip(x) and len(layers) >= 2 \
+    if use_hip(x) and len(layers) >= 2 and _COMPUTE_DTYPE == "fp32" \
             and all(m.ln_weight is not None for m in layers):
         acts = tuple(_ACT_CODES[m.act] for m in layers)
         params = []

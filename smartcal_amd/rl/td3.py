@@ -109,6 +109,19 @@ class Agent:
         self.time_step += 1
         return mu_prime.cpu().numpy()
 
+    def choose_action_tensor(self, observation):
+        """Device-resident action (no host sync) — the bench/vectorized
+        rollout path; ``choose_action`` keeps the reference's numpy API."""
+        if self.time_step < self.warmup:
+            mu = torch.randn(self.n_actions, device=self.device) * self.noise
+        else:
+            state = obs_to_state(observation).to(self.device)
+            with torch.no_grad():
+                mu = self.actor(state).reshape(-1)
+        noise = torch.randn(self.n_actions, device=self.device) * self.noise
+        self.time_step += 1
+        return (mu + noise).clamp(self.min_action, self.max_action)
+
     def _grad_sync(self, fps):
         if self.grad_hook is not None:
             self.grad_hook(fps)
