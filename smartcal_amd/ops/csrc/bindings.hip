@@ -64,6 +64,13 @@ extern "C" __global__ void gemm_bf16_nn_kernel(const float*, const float*,
                                                float*, int, int, int);
 extern "C" __global__ void gemm_bf16_tn_kernel(const float*, const float*,
                                                float*, int, int, int, int);
+extern "C" __global__ void attn_fwd_kernel(const float*, const float*,
+                                           const float*, float*, float*,
+                                           int, int, int);
+extern "C" __global__ void attn_bwd_kernel(const float*, const float*,
+                                           const float*, const float*,
+                                           const float*, float*, float*,
+                                           float*, int, int, int);
 extern "C" __global__ void per_update_kernel(float*, const long*,
                                              const float*, int, float, float,
                                              float);
@@ -549,6 +556,50 @@ void per_update(at::Tensor& priorities, const at::Tensor& idx,
                      (float)eps, (float)alpha, (float)max_priority);
 }
 
+// Fused small-sequence attention: O = softmax(Q K^T / sqrt(dh)) V in one
+// launch per call; A (softmax) returned for backward/inspection.
+std::tuple<at::Tensor, at::Tensor> attn_fwd(const at::Tensor& Q,
+                                            const at::Tensor& K,
+                                            const at::Tensor& V) {
+  check_f32(Q, "Q");
+  check_f32(K, "K");
+  check_f32(V, "V");
+  const int G = Q.size(0), T = Q.size(1), dh = Q.size(2);
+  TORCH_CHECK(T <= 32 && dh <= 96,
+              "attn_fwd caps: T <= 32, dh <= 96 (python falls back)");
+  auto O = at::empty_like(Q);
+  auto A = at::empty({G, T, T}, Q.options());
+  const int Tp = (T + 15) & ~15;
+  const int dp = ((dh + 3) & ~3) + 1;
+  const size_t lds = (size_t)(3 * Tp * dp + Tp * (Tp + 1)) * sizeof(float);
+  hipLaunchKernelGGL(attn_fwd_kernel, dim3(G), dim3(64), lds, stream(),
+                     Q.data_ptr<float>(), K.data_ptr<float>(),
+                     V.data_ptr<float>(), O.data_ptr<float>(),
+                     A.data_ptr<float>(), G, T, dh);
+  return {O, A};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> attn_bwd(
+    const at::Tensor& dO, const at::Tensor& A, const at::Tensor& Q,
+    const at::Tensor& K, const at::Tensor& V) {
+  check_f32(dO, "dO");
+  check_f32(A, "A");
+  const int G = Q.size(0), T = Q.size(1), dh = Q.size(2);
+  auto dQ = at::empty_like(Q);
+  auto dK = at::empty_like(Q);
+  auto dV = at::empty_like(Q);
+  const int Tp = (T + 15) & ~15;
+  const int dp = ((dh + 3) & ~3) + 1;
+  const size_t lds =
+      (size_t)(4 * Tp * dp + 2 * Tp * (Tp + 1) + Tp) * sizeof(float);
+  hipLaunchKernelGGL(attn_bwd_kernel, dim3(G), dim3(64), lds, stream(),
+                     dO.data_ptr<float>(), A.data_ptr<float>(),
+                     Q.data_ptr<float>(), K.data_ptr<float>(),
+                     V.data_ptr<float>(), dQ.data_ptr<float>(),
+                     dK.data_ptr<float>(), dV.data_ptr<float>(), G, T, dh);
+  return {dQ, dK, dV};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -572,4 +623,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_linear_bf16_fwd", &fused_linear_bf16_fwd);
   m.def("mfma_gemm_nn_bf16", &mfma_gemm_nn_bf16);
   m.def("mfma_gemm_tn_bias_into_bf16", &mfma_gemm_tn_bias_into_bf16);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
 }
