@@ -28,6 +28,7 @@ sources = [str(CSRC / f) for f in [
     "conv2d.hip",
     "per.hip",
     "fused_linear_bf16.hip",
+    "attention.hip",
 ]]
 
 setup(

@@ -71,10 +71,11 @@ class SupervisedBuffer:
 
 
 def scaled_dot_product(q, k, v):
-    d_k = q.size()[-1]
-    attn_logits = torch.matmul(q, k.transpose(-2, -1)) / math.sqrt(d_k)
-    attention = F.softmax(attn_logits, dim=-1)
-    return torch.matmul(attention, v), attention
+    """One fused HIP launch on GPU (ops/csrc/attention.hip: LDS-staged
+    MFMA tiles + in-LDS softmax); torch composition on CPU (the oracle).
+    Same contract as the reference's (`transformer_models.py:76-83`)."""
+    from ..ops.attention import scaled_dot_product as _sdp
+    return _sdp(q, k, v)
 
 
 class MultiheadAttention(nn.Module):

@@ -53,13 +53,14 @@ class TokenMHA(nn.Module):
         self.o.bias.data.fill_(0)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.attention import scaled_dot_product as _sdp
         B, T, d = x.shape
         qkv = self.qkv(x.reshape(B * T, d)).reshape(B, T, 3, self.h,
                                                     self.dh)
         q, k, v = qkv.permute(2, 0, 3, 1, 4)          # (B, h, T, dh) each
-        att = torch.softmax(q @ k.transpose(-2, -1) / math.sqrt(self.dh),
-                            dim=-1)
-        out = (att @ v).transpose(1, 2).reshape(B * T, d)
+        # fused attention kernel on GPU (one launch per call)
+        vals, _ = _sdp(q.contiguous(), k.contiguous(), v.contiguous())
+        out = vals.transpose(1, 2).reshape(B * T, d)
         return self.o(out).reshape(B, T, d)
 
 
