@@ -29,6 +29,7 @@ sources = [str(CSRC / f) for f in [
     "per.hip",
     "fused_linear_bf16.hip",
     "attention.hip",
+    "cgemm.hip",
 ]]
 
 setup(

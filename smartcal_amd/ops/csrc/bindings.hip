@@ -71,6 +71,9 @@ extern "C" __global__ void attn_bwd_kernel(const float*, const float*,
                                            const float*, const float*,
                                            const float*, float*, float*,
                                            float*, int, int, int);
+extern "C" __global__ void cgemm_nn_bcast_kernel(const float*, const float*,
+                                                 float*, int, int, int, int,
+                                                 int);
 extern "C" __global__ void per_update_kernel(float*, const long*,
                                              const float*, int, float, float,
                                              float);
@@ -556,6 +559,28 @@ void per_update(at::Tensor& priorities, const at::Tensor& idx,
                      (float)eps, (float)alpha, (float)max_priority);
 }
 
+// Batched complex64 GEMM with A broadcast over `rep` consecutive batch
+// elements: C[g] = A[g / rep] @ B[g]. The dsolutions hot op.
+at::Tensor cgemm_nn_bcast(const at::Tensor& A, const at::Tensor& B,
+                          int64_t rep) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kComplexFloat
+              && A.is_contiguous(), "A must be contiguous cfloat GPU");
+  TORCH_CHECK(B.is_cuda() && B.scalar_type() == at::kComplexFloat
+              && B.is_contiguous(), "B must be contiguous cfloat GPU");
+  const int KA = A.size(0), M = A.size(1), Kd = A.size(2);
+  const int G = B.size(0), N = B.size(2);
+  TORCH_CHECK(B.size(1) == Kd, "cgemm K mismatch");
+  TORCH_CHECK(G == KA * rep, "cgemm batch/rep mismatch");
+  auto C = at::empty({G, M, N}, A.options());
+  dim3 grid((M + 63) / 64, (N + 15) / 16, G);
+  hipLaunchKernelGGL(cgemm_nn_bcast_kernel, grid, dim3(256), 0, stream(),
+                     reinterpret_cast<const float*>(A.data_ptr()),
+                     reinterpret_cast<const float*>(B.data_ptr()),
+                     reinterpret_cast<float*>(C.data_ptr()),
+                     M, Kd, N, G, (int)rep);
+  return C;
+}
+
 // Fused small-sequence attention: O = softmax(Q K^T / sqrt(dh)) V in one
 // launch per call; A (softmax) returned for backward/inspection.
 std::tuple<at::Tensor, at::Tensor> attn_fwd(const at::Tensor& Q,
@@ -625,4 +650,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_gemm_tn_bias_into_bf16", &mfma_gemm_tn_bias_into_bf16);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("cgemm_nn_bcast", &cgemm_nn_bcast);
 }

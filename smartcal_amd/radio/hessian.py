@@ -145,8 +145,19 @@ def dsolutions_r(C: torch.Tensor, J: torch.Tensor, N: int,
     # inv+GEMM instead of batched solve: ROCm's hipblasCtrsmBatched fails
     # for complex64 with nrhs ≥ 1024 (B can be 1891 at LOFAR scale), and
     # the GEMM path keeps MFMA busy; Dgrad is ε-regularized
-    Ainv = torch.linalg.inv(Dgrad + _EPS * eye).unsqueeze(1)  # (K,1,4N,4N)
-    dJ = Ainv @ AdV.permute(1, 0, 2, 3)                       # (K,8,4N,B)
+    Ainv = torch.linalg.inv(Dgrad + _EPS * eye)               # (K,4N,4N)
+    from ..ops import use_hip
+    if use_hip(AdV.real):
+        # hand-written batched complex MFMA GEMM (ops/csrc/cgemm.hip):
+        # rocBLAS runs this complex shape at ~1.1 TF/s — the single
+        # biggest cost of the influence pipeline before this kernel
+        from ..ops import ext
+        Bv = AdV.permute(1, 0, 2, 3).contiguous() \
+                .reshape(K * 8, 4 * N, B)                     # (K*8,4N,B)
+        dJ = ext().cgemm_nn_bcast(Ainv.contiguous(), Bv, 8)
+        dJ = dJ.reshape(K, 8, 4 * N, B)
+    else:
+        dJ = Ainv.unsqueeze(1) @ AdV.permute(1, 0, 2, 3)      # (K,8,4N,B)
     return dJ.permute(1, 0, 2, 3).contiguous()
 
 
