@@ -57,6 +57,10 @@ def main():
         timeit(lambda: hs.dresiduals_r(C, J, N, dJ, False)), 2)
     out["dresiduals_rk_ms"] = round(
         timeit(lambda: hs.dresiduals_rk(C, J, N, dJ, False)), 2)
+    out["dres_colmeans_ms"] = round(
+        timeit(lambda: hs.dres_colmeans(C, J, N, H)), 2)
+    out["dres_colmeans_perk_ms"] = round(
+        timeit(lambda: hs.dres_colmeans(C, J, N, H, per_k=True)), 2)
     # the dsolutions GEMM alone, old path vs kernel
     eye = torch.eye(4 * N, dtype=C.dtype, device=DEV)
     Ainv = torch.linalg.inv(H + 1e-12 * eye)

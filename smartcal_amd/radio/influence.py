@@ -67,12 +67,9 @@ def influence_values(residual4: torch.Tensor, C: torch.Tensor,
         H = hs.hessianres(Rchunk, Cchunk, Jchunk, N)
         if Hadd is not None:
             H = H + Hadd
-        dJ = hs.dsolutions_r(Cchunk, Jchunk, N, H)
-        dR = hs.dresiduals_r(Cchunk, Jchunk, N, dJ, False)  # (8,4B,B)
-        # mean over the B row-blocks for each polarization offset,
-        # summed over the 8 perturbation directions → (B,) per pol
-        dRb = dR.reshape(8, B, 4, B)
-        m = dRb.mean(dim=1)                                  # (8,4,B)
+        # analytic row-block means (hs.dres_colmeans): never builds the
+        # (8,4B,B) dR nor the 8·K·4N·B dJ — a 4-RHS solve + contractions
+        m = hs.dres_colmeans(Cchunk, Jchunk, N, H)           # (8,4,B)
         xx = m[:, 0].sum(dim=0)
         yy = m[:, 3].sum(dim=0)
         out[s0:s1, 0] = xx.repeat(Tdelta)
@@ -106,9 +103,8 @@ def influence_per_direction(residual4: torch.Tensor, C: torch.Tensor,
         H = hs.hessianres(Rchunk, Cchunk, Jchunk, N)
         if Hadd is not None:
             H = H + Hadd
-        dJ = hs.dsolutions_r(Cchunk, Jchunk, N, H)
-        dRk = hs.dresiduals_rk(Cchunk, Jchunk, N, dJ, False)  # (8,K,4B,B)
-        m = dRk.reshape(8, K, B, 4, B).mean(dim=2)            # (8,K,4,B)
+        m = hs.dres_colmeans(Cchunk, Jchunk, N, H,
+                             per_k=True)                       # (8,K,4,B)
         ms = m.sum(dim=0)                                      # (K,4,B)
         for pol in (0, 3):
             out[:, s0:s1, pol] = ms[:, pol, :].repeat(1, Tdelta)

@@ -312,3 +312,123 @@ def test_learner_two_actors():
     scores = q.get()
     assert len(scores) == 2
     assert all(np.isfinite(s) for s in scores)
+
+
+# ---- world-8 coverage (the driver's 8-GPU shapes, on CPU/gloo) ----
+
+def _la8_worker(rank, world, port, q):
+    from smartcal_amd.distributed.learner_actor import run_process
+    from smartcal_amd.envs.enet import ENetEnv
+    from smartcal_amd.rl.sac import Agent
+
+    torch.manual_seed(rank)
+    np.random.seed(rank)
+
+    def agent_factory():
+        return Agent(gamma=0.99, batch_size=4, n_actions=NACT, tau=0.005,
+                     max_mem_size=128, input_dims=[OBS_DIM], lr_a=1e-3,
+                     lr_c=1e-3, reward_scale=N, alpha=0.03,
+                     device=torch.device("cpu"))
+
+    def env_factory():
+        return ENetEnv(M, N, device=torch.device("cpu"))
+
+    scores = run_process(rank, world, agent_factory, env_factory,
+                         obs_dim=OBS_DIM, n_actions=NACT, episodes=1,
+                         epochs=1, steps=2, learner_addr="127.0.0.1",
+                         learner_port=port, max_transitions=4,
+                         backend="gloo")
+    if rank == 0:
+        q.put(scores)
+
+
+@pytest.mark.timeout(600)
+def test_learner_seven_actors_world8():
+    """BASELINE config 4 shape: 1 learner + 7 actors (world 8), one
+    episode round — weight broadcast + 7-way transition gather."""
+    port = 29561
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_la8_worker, args=(r, 8, port, q))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(540)
+    assert all(p.exitcode == 0 for p in procs), \
+        [p.exitcode for p in procs]
+    scores = q.get()
+    assert len(scores) == 1 and np.isfinite(scores[0])
+
+
+@pytest.mark.timeout(900)
+def test_bench_torchrun_world8_cpu(tmp_path):
+    """The driver's 8-GPU bench launch (BASELINE config 4/5 shape) on
+    CPU/gloo: 8 ranks, DP grad all-reduce, one JSON line from rank 0."""
+    import json as _json
+    import subprocess, os, sys
+    from pathlib import Path
+    ROOT = Path(__file__).resolve().parents[1]
+    env = dict(os.environ, PYTHONPATH=str(ROOT))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29518", str(ROOT / "bench.py"),
+         "--gpus", "8", "--steps", "2", "--warmup", "0"],
+        cwd=tmp_path, env=env, capture_output=True, text=True, timeout=840)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = _json.loads(line)
+    assert out["n_gpus"] == 8
+    assert out["config"]["parallelism"] == "dp8"
+    assert out["value"] > 0
+
+
+def _solver8_worker(rank, world, port, q):
+    import torch.distributed as dist
+    from smartcal_amd.radio import array as arr, sim, solver
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    rng = np.random.default_rng(3)     # same scenario on all ranks
+    layout = arr.lofar_like_layout(N=6, rng=rng)
+    sky, cs, *_, ra0, dec0 = sim.make_demixing_sky(rng, n_outliers=1)
+    freqs_all = np.linspace(120e6, 160e6, 8)
+    vis = sim.simulate_observation(layout, sky, cs, freqs_all, ra0, dec0,
+                                   Ts=1, Tdelta=4, snr=50.0, rng=rng,
+                                   torch_seed=0)
+    mine = slice(rank, rank + 1)       # one frequency per rank
+    shard = sim.VisData(uvw=vis.uvw, freqs=vis.freqs[mine],
+                        data=vis.data[mine], N=vis.N, ra0=ra0, dec0=dec0,
+                        Ts=1, Tdelta=4)
+    sol = solver.calibrate(shard, sky, cs, np.full(2, 5.0, np.float32),
+                           admm_iter=2, poly_order=2, n_sweeps=1,
+                           init_sweeps=3)
+    q.put((rank, sol.Z.cpu()))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_solver_freq_sharded_world8():
+    """Consensus-ADMM with 8 frequency shards at world 8 — the full
+    8-GPU Z-exchange shape (BASELINE docal.sh topology, one band per
+    rank). All ranks must agree on the global Z."""
+    port = 29563
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_solver8_worker, args=(r, 8, port, q))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    out = {}
+    for _ in range(8):
+        r, Z = q.get()
+        out[r] = Z
+    for p in procs:
+        p.join(540)
+    assert all(p.exitcode == 0 for p in procs), \
+        [p.exitcode for p in procs]
+    for r in range(1, 8):
+        torch.testing.assert_close(out[0], out[r], rtol=1e-4, atol=1e-5)

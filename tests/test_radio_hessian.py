@@ -176,3 +176,36 @@ def test_llr(prob):
     want = _oracle_llr(R, C, J, N)
     got = hs.log_likelihood_ratio(Rt, Ct, Jt, N).numpy()
     np.testing.assert_allclose(got, want, rtol=1e-4)
+
+
+def test_dres_colmeans_matches_composed_path():
+    """Analytic row-block means (4-RHS solve + contractions) vs the full
+    dsolutions_r -> dresiduals_r/_rk -> reshape/mean composition."""
+    import numpy as np
+    import torch
+    from smartcal_amd.radio import hessian as hs
+    rng = np.random.default_rng(0)
+    N, T, K = 8, 3, 3
+    B = N * (N - 1) // 2
+    S = B * T
+    C = torch.from_numpy((rng.standard_normal((K, S, 4))
+                          + 1j * rng.standard_normal((K, S, 4))
+                          ).astype(np.complex64))
+    J = torch.from_numpy((rng.standard_normal((K, 2 * N, 2))
+                          + 1j * rng.standard_normal((K, 2 * N, 2))
+                          ).astype(np.complex64))
+    R = torch.from_numpy((rng.standard_normal((2 * S, 2))
+                          + 1j * rng.standard_normal((2 * S, 2))
+                          ).astype(np.complex64))
+    H = hs.hessianres(R, C, J, N)
+    dJ = hs.dsolutions_r(C, J, N, H)
+    m_ref = hs.dresiduals_r(C, J, N, dJ, False) \
+        .reshape(8, B, 4, B).mean(dim=1)
+    m_new = hs.dres_colmeans(C, J, N, H)
+    assert float((m_new - m_ref).abs().max()
+                 / m_ref.abs().max()) < 1e-5
+    mk_ref = hs.dresiduals_rk(C, J, N, dJ, False) \
+        .reshape(8, K, B, 4, B).mean(dim=2)
+    mk_new = hs.dres_colmeans(C, J, N, H, per_k=True)
+    assert float((mk_new - mk_ref).abs().max()
+                 / mk_ref.abs().max()) < 1e-5
