@@ -31,7 +31,7 @@ from .io import load_visdata, save_visdata
 from .sim import VisData
 
 __all__ = ["read_ms", "ms_to_npz", "observation_from_npz",
-           "merge_visdata"]
+           "merge_visdata", "average_visdata"]
 
 
 def _casacore_tables():
@@ -107,6 +107,37 @@ def merge_visdata(parts: Sequence[VisData]) -> VisData:
                    data=data[list(order)], N=v0.N, ra0=v0.ra0,
                    dec0=v0.dec0, Ts=v0.Ts, Tdelta=v0.Tdelta,
                    noise_sigma=v0.noise_sigma)
+
+
+def average_visdata(vis: VisData, time_factor: int = 1,
+                    freq_factor: int = 1) -> VisData:
+    """Time/frequency averaging of an observation — the in-memory
+    equivalent of the DP3 averaging step in the reference's
+    ``extract_dataset`` (`generate_data.py:623-694`: real MS → averaged
+    working set). Visibilities are averaged over ``time_factor``
+    consecutive timeslots (uvw taken from the first slot of each group)
+    and ``freq_factor`` consecutive bands (frequencies averaged).
+    """
+    B = vis.B
+    T = vis.n_time
+    Nf = vis.data.shape[0]
+    tf = max(int(time_factor), 1)
+    ff = max(int(freq_factor), 1)
+    Tn = T // tf
+    Fn = Nf // ff
+    if Tn < 1 or Fn < 1:
+        raise ValueError("averaging factors exceed the data extent")
+    d = vis.data[:Fn * ff, :Tn * tf * B]
+    d = d.reshape(Fn, ff, Tn, tf, B, 4).mean(dim=(1, 3))
+    uvw = vis.uvw[:Tn * tf * B].reshape(Tn, tf, B, 3)[:, 0] \
+        .reshape(Tn * B, 3)
+    freqs = np.asarray(vis.freqs).reshape(-1)[:Fn * ff] \
+        .reshape(Fn, ff).mean(axis=1)
+    return VisData(uvw=uvw, freqs=freqs,
+                   data=d.reshape(Fn, Tn * B, 4), N=vis.N, ra0=vis.ra0,
+                   dec0=vis.dec0, Ts=max(vis.Ts // tf, 1),
+                   Tdelta=max(vis.Tdelta // tf, 1),
+                   noise_sigma=vis.noise_sigma)
 
 
 def observation_from_npz(paths: Sequence[str], device="cpu") -> VisData:

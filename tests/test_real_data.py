@@ -127,3 +127,20 @@ def test_read_ms_raises_helpfully_without_casacore():
         pass
     with pytest.raises(ImportError, match="ms_to_npz"):
         ms_io.read_ms("/nonexistent.ms")
+
+
+def test_average_visdata_time_freq(tmp_path):
+    """DP3-style averaging (extract_dataset equivalent): means check out
+    and geometry fields survive."""
+    from smartcal_amd.radio import ms_io
+    paths, vis = _fixture_obs(tmp_path, Nf=2)
+    merged = ms_io.observation_from_npz(paths)
+    av = ms_io.average_visdata(merged, time_factor=2, freq_factor=2)
+    B = vis.B
+    assert av.data.shape[0] == 1
+    assert av.n_time == vis.n_time // 2
+    # first averaged sample = mean over (2 slots x 2 bands) of sample 0
+    expect = (vis.data[0, 0] + vis.data[0, B]
+              + vis.data[1, 0] + vis.data[1, B]) / 4
+    torch.testing.assert_close(av.data[0, 0], expect)
+    assert np.isclose(av.freqs[0], np.mean(vis.freqs))
