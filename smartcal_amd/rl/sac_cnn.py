@@ -213,6 +213,8 @@ class Agent:
     def learn(self):
         if len(self.replaymem) < self.batch_size:
             return
+        if getattr(self, "_graph", None) is not None:
+            return self._learn_graphed()
         if self.prioritized:
             batch, idx, is_w = self.replaymem.sample_buffer(self.batch_size)
             is_w = is_w.unsqueeze(1)
@@ -220,9 +222,11 @@ class Agent:
             batch = self.replaymem.sample_buffer(self.batch_size)
             idx = is_w = None
         (img, meta, action, reward, img_, meta_, done, hint) = batch
-        reward = reward.unsqueeze(1)
-        done = done.unsqueeze(1)
+        self._learn_body(img, meta, action, reward.unsqueeze(1),
+                         img_, meta_, done.unsqueeze(1), hint, is_w, idx)
 
+    def _learn_body(self, img, meta, action, reward, img_, meta_, done,
+                    hint, is_w=None, idx=None):
         with torch.no_grad():
             na, nlp = self.actor.sample_normal(img_, meta_,
                                                reparameterize=False)
@@ -295,6 +299,76 @@ class Agent:
                 self.rho = self.rho + self.admm_rho * gfun
         self.learn_counter += 1
         self.update_network_parameters()
+
+    # -- hipGraph capture of the CNN learn step -----------------------------
+    def enable_cuda_graph(self):
+        """Capture the whole CNN learn step (conv forwards/backwards on
+        the direct-conv kernels, BatchNorm batch stats + running-stat
+        updates — all static-shape tensor ops — fused Adam, polyak) into
+        ONE hipGraph. Plain-replay, no-hint, fixed-alpha path only."""
+        assert not self.use_hint and not self.prioritized \
+            and not self.learn_alpha
+        assert self.device.type == "cuda"
+        B = self.batch_size
+        mem = self.replaymem
+        dev = self.device
+        self._g_idx = torch.zeros(B, dtype=torch.long, device=dev)
+        self._g_img = torch.zeros((B, *mem.img_memory.shape[1:]),
+                                  device=dev)
+        self._g_img_ = torch.zeros_like(self._g_img)
+        self._g_meta = torch.zeros((B, *mem.meta_memory.shape[1:]),
+                                   device=dev)
+        self._g_meta_ = torch.zeros_like(self._g_meta)
+        self._g_action = torch.zeros(B, self.n_actions, device=dev)
+        self._g_reward = torch.zeros(B, 1, device=dev)
+        self._g_done = torch.zeros(B, 1, dtype=torch.bool, device=dev)
+        self._g_hint = torch.zeros(B, self.n_actions, device=dev)
+        self._fill_static()
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                self._learn_body(self._g_img, self._g_meta, self._g_action,
+                                 self._g_reward, self._g_img_,
+                                 self._g_meta_, self._g_done, self._g_hint)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._learn_body(self._g_img, self._g_meta, self._g_action,
+                             self._g_reward, self._g_img_, self._g_meta_,
+                             self._g_done, self._g_hint)
+        self._graph = g
+
+    def disable_cuda_graph(self):
+        self._graph = None
+
+    def _fill_static(self):
+        mem = self.replaymem
+        n = len(mem)
+        torch.randint(0, n, (self.batch_size,), device=self.device,
+                      out=self._g_idx)
+        torch.index_select(mem.img_memory, 0, self._g_idx,
+                           out=self._g_img)
+        torch.index_select(mem.new_img_memory, 0, self._g_idx,
+                           out=self._g_img_)
+        torch.index_select(mem.meta_memory, 0, self._g_idx,
+                           out=self._g_meta)
+        torch.index_select(mem.new_meta_memory, 0, self._g_idx,
+                           out=self._g_meta_)
+        torch.index_select(mem.action_memory, 0, self._g_idx,
+                           out=self._g_action)
+        torch.index_select(mem.reward_memory, 0, self._g_idx,
+                           out=self._g_reward.view(-1))
+        torch.index_select(mem.terminal_memory, 0, self._g_idx,
+                           out=self._g_done.view(-1))
+
+    def _learn_graphed(self):
+        self._fill_static()
+        self._graph.replay()
+        # the captured body's python-side counter bump does not replay
+        self.learn_counter += 1
 
     # -- checkpointing (reference naming: <prefix>_sac_{actor,critic}) ----
     def _path(self, name):

