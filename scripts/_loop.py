@@ -13,12 +13,29 @@ import pickle
 import numpy as np
 
 
+def _try_enable_graph(agent):
+    """Capture the learn step into one hipGraph when the agent/config
+    supports it (GPU, plain replay, no hint/learnable-alpha). Safe no-op
+    otherwise."""
+    if getattr(agent, "_graph", None) is not None:
+        return True
+    enable = getattr(agent, "enable_cuda_graph", None)
+    if enable is None:
+        return False
+    try:
+        enable()
+        return True
+    except (AssertionError, RuntimeError, AttributeError):
+        return False
+
+
 def run_training(env, agent, episodes: int, steps: int,
                  provide_hint: bool = False, save_every: int = 10,
                  scores_file: str = "scores.pkl",
                  reward_shaping=None, warmup_episodes: int = 0,
-                 verbose: bool = True):
+                 verbose: bool = True, use_graph: bool = True):
     scores = []
+    graph_tried = False
     for i in range(episodes):
         score = 0.0
         done = False
@@ -39,6 +56,11 @@ def run_training(env, agent, episodes: int, steps: int,
             agent.store_transition(observation, action, reward,
                                    observation_, done, hint)
             score += float(reward)
+            if use_graph and not graph_tried \
+                    and len(getattr(agent, "replaymem", ())) \
+                    >= getattr(agent, "batch_size", 1):
+                graph_tried = True
+                _try_enable_graph(agent)
             agent.learn()
             observation = observation_
             loop += 1
