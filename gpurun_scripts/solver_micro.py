@@ -34,6 +34,6 @@ for E in [1, 8, 64, 256]:
         print(f"E={E:4d} epochs={ep:2d}: solve {us:9.1f} us "
               f"({us/E:7.2f} us/env)")
     x, Y, S, nh = ops.ext().enet_lbfgs_solve(A, y, rho, 20, 10, 7)
-    us = timeit(lambda: ops.ext().enet_influence(A, y, x, Y, S, nh, pen))
+    us = timeit(lambda: ops.ext().enet_influence(A, y, x, Y, S, nh, pen, rho))
     print(f"E={E:4d} influence: {us:9.1f} us ({us/E:7.2f} us/env)  "
           f"nhist={int(nh[0])}")

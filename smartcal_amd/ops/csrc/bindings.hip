@@ -53,7 +53,7 @@ extern "C" __global__ void als_sweep_kernel(
     const int*, c32b*, c32b*, int, int, int, int, int, int);
 extern "C" __global__ void enet_influence_kernel(
     const float*, const float*, const float*, const float*, const float*,
-    const int*, const float*, float*, float*, int, int, int);
+    const int*, const float*, const float*, float*, float*, int, int, int);
 extern "C" __global__ void per_sample_kernel(const float*, const float*,
                                              long*, float*, float*, int, int,
                                              float);
@@ -500,10 +500,12 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> enet_lbfgs_solve(
 std::tuple<at::Tensor, at::Tensor> enet_influence(
     const at::Tensor& A, const at::Tensor& y, const at::Tensor& x,
     const at::Tensor& Y, const at::Tensor& S, const at::Tensor& nh,
-    const at::Tensor& penalty) {
+    const at::Tensor& penalty, const at::Tensor& rho) {
   check_f32(A, "A");
+  check_f32(rho, "rho");
   const int E = A.size(0), N = A.size(1), M = A.size(2);
   TORCH_CHECK(N <= 32 && M <= 32, "enet influence supports N,M <= 32");
+  TORCH_CHECK(rho.numel() == 2 * E, "rho must be (E, 2)");
   auto EE = at::empty({E, N}, A.options());
   auto reward = at::empty({E}, A.options());
   const int lds_floats = N * M + M * N + N * N + 2 * HMAX * M + HMAX * N + N
@@ -513,8 +515,9 @@ std::tuple<at::Tensor, at::Tensor> enet_influence(
                      A.data_ptr<float>(), y.data_ptr<float>(),
                      x.data_ptr<float>(), Y.data_ptr<float>(),
                      S.data_ptr<float>(), nh.data_ptr<int>(),
-                     penalty.data_ptr<float>(), EE.data_ptr<float>(),
-                     reward.data_ptr<float>(), E, N, M);
+                     penalty.data_ptr<float>(), rho.data_ptr<float>(),
+                     EE.data_ptr<float>(), reward.data_ptr<float>(),
+                     E, N, M);
   return {EE, reward};
 }
 

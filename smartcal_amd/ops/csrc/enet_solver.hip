@@ -468,6 +468,7 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
     const float* __restrict__ Sg,   // (E, HMAX, M)
     const int* __restrict__ nhistg, // (E,)
     const float* __restrict__ peng, // (E,)
+    const float* __restrict__ rhog, // (E, 2) — rho1 drives the filter
     float* __restrict__ EEg,        // (E, N)
     float* __restrict__ rewardg,    // (E,)
     int E, int N, int M) {
@@ -506,17 +507,22 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
   for (int m = 0; m < 32; ++m)
     qc[m] = (lane < N && m < M) ? -2.f * A[lane * M + m] : 0.f;
 
-  // Degenerate-pair filter: the true Hessian is 2(A^T A + rho1 I), so
-  // ys/ss >= 2 rho1 for any genuine curvature pair; pairs far below that
-  // are line-search noise whose 1/ys blows the two-loop up (observed as
-  // ~1e6-magnitude eigenvalue-ratio rewards). ro = 0 no-ops a pair in
-  // both loops. The CPU oracle (`ops/enet.py::influence_eigs_reference`)
-  // applies the same filter.
+  // Degenerate-pair filter AT THE ANALYTIC BOUND: the true Hessian is
+  // 2(A^T A + rho1 I) and the L1 subgradient difference is monotone, so
+  // every genuine curvature pair has ys >= 2 rho1 ss. Pairs below
+  // rho1 ss (half the floor) are line-search noise whose 1/ys blows the
+  // two-loop up — with the old fixed 1e-6 cutoff, noise pairs in
+  // (1e-6, rho1) ss still inflated H^-1 ~1000x over the worst genuine
+  // pair and produced the huge negative min(EE)/max(EE) rewards the
+  // round-2 curve audit caught. ro = 0 no-ops a pair in both loops. The
+  // CPU oracle (`ops/enet.py::influence_eigs_reference`) applies the
+  // same filter.
+  const float thr = fmaxf(1e-6f, rhog[env * 2]);
   float ys = 1.f, yy = 1.f;
   for (int i = 0; i < nh; ++i) {
     float ysi = lds_dot(&Yv[i * M], &Sv[i * M], M);
     float ssi = lds_dot(&Sv[i * M], &Sv[i * M], M);
-    float r = (ysi > 1e-6f * ssi) ? 1.f / ysi : 0.f;
+    float r = (ysi > thr * ssi) ? 1.f / ysi : 0.f;
     if (lane == 0) ro[i] = r;
     if (r != 0.f) {  // H_diag scale from the newest GOOD pair
       ys = ysi;

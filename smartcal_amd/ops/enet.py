@@ -73,7 +73,8 @@ def curvature_stacks(opt) -> Tuple[torch.Tensor, torch.Tensor]:
 
 
 def influence_eigs_reference(A: torch.Tensor, Y: torch.Tensor,
-                             S: torch.Tensor) -> torch.Tensor:
+                             S: torch.Tensor,
+                             rho1: float = 0.0) -> torch.Tensor:
     """EE = 1 + eigvals(A · H^{-1} · (-2 A^T)) via the matrix two-loop.
 
     Column-identical to the reference's per-column ``inv_hessian_mult`` loop
@@ -82,13 +83,14 @@ def influence_eigs_reference(A: torch.Tensor, Y: torch.Tensor,
     from ..autograd_tools import inv_hessian_mult_mat
 
     if Y.shape[0] > 0:
-        # degenerate-pair filter (same as the HIP influence kernel): the
-        # true Hessian 2(A^T A + rho1 I) bounds ys/ss >= 2 rho1, so pairs
-        # far below are line-search noise whose 1/ys explodes the
-        # two-loop (and with it the min(EE)/max(EE) reward term)
+        # degenerate-pair filter at the analytic bound (same as the HIP
+        # influence kernel): the true Hessian 2(A^T A + rho1 I) and the
+        # monotone L1 subgradient give ys >= 2 rho1 ss for every genuine
+        # pair; pairs below rho1 ss are line-search noise whose 1/ys
+        # explodes the two-loop (and the min(EE)/max(EE) reward term)
         ys = (Y * S).sum(-1)
         ss = (S * S).sum(-1)
-        good = ys > 1e-6 * ss
+        good = ys > max(1e-6, float(rho1)) * ss
         Y, S = Y[good], S[good]
     Q = -2.0 * A.t().contiguous()
     mm = inv_hessian_mult_mat(Y, S, Q)
@@ -107,11 +109,12 @@ def solve_and_influence_device(A: torch.Tensor, y: torch.Tensor,
     ext()  # loud failure if the extension is missing on a GPU box
     Ab = A.unsqueeze(0).contiguous()
     yb = y.unsqueeze(0).contiguous()
-    x, Yc, Sc, nh = ext().enet_lbfgs_solve(Ab, yb,
-                                           rho.reshape(1, 2).contiguous(),
+    rho2d = rho.reshape(1, 2).contiguous()
+    x, Yc, Sc, nh = ext().enet_lbfgs_solve(Ab, yb, rho2d,
                                            epochs, max_iter, history)
     EE, reward = ext().enet_influence(Ab, yb, x, Yc, Sc, nh,
-                                      penalty.reshape(1).contiguous())
+                                      penalty.reshape(1).contiguous(),
+                                      rho2d)
     return x[0], EE[0], reward[0]
 
 
@@ -133,13 +136,13 @@ def solve_and_influence(A: torch.Tensor, y: torch.Tensor,
         x, Yc, Sc, nh = ext().enet_lbfgs_solve(Ab, yb, rho, epochs,
                                                max_iter, history)
         pen = torch.tensor([penalty], device=A.device)
-        EE, reward = ext().enet_influence(Ab, yb, x, Yc, Sc, nh, pen)
+        EE, reward = ext().enet_influence(Ab, yb, x, Yc, Sc, nh, pen, rho)
         return x[0], EE[0], reward[0]
 
     x, opt = lbfgs_solve_reference(A, y, float(rho1), float(rho2),
                                    epochs, max_iter, history)
     Y, S = curvature_stacks(opt)
-    EE = influence_eigs_reference(A, Y, S)
+    EE = influence_eigs_reference(A, Y, S, rho1=float(rho1))
     final_err = torch.norm(A @ x - y, 2)
     reward = torch.norm(y, 2) / final_err + EE.min() / EE.max() + penalty
     return x, EE, reward
@@ -162,10 +165,11 @@ def solve_and_influence_batch(A: torch.Tensor, y: torch.Tensor,
     if use_hip(A):
         Ab = A.contiguous()
         yb = y.contiguous()
-        x, Yc, Sc, nh = ext().enet_lbfgs_solve(Ab, yb, rho.contiguous(),
+        rhoc = rho.contiguous()
+        x, Yc, Sc, nh = ext().enet_lbfgs_solve(Ab, yb, rhoc,
                                                epochs, max_iter, history)
         EE, reward = ext().enet_influence(Ab, yb, x, Yc, Sc, nh,
-                                          penalty.contiguous())
+                                          penalty.contiguous(), rhoc)
         return x, EE, reward
     xs, EEs, rs = [], [], []
     for e in range(A.shape[0]):

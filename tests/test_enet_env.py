@@ -51,7 +51,7 @@ def test_influence_eigs_vs_reference_loop():
     y = torch.randn(N)
     x, opt = enet_ops.lbfgs_solve_reference(A, y, 0.05, 0.01, epochs=10)
     Y, S = enet_ops.curvature_stacks(opt)
-    EE_fast = enet_ops.influence_eigs_reference(A, Y, S)
+    EE_fast = enet_ops.influence_eigs_reference(A, Y, S, rho1=0.05)
 
     # reference formulation: per-column inv_hessian_mult + torch.linalg.eig
     ll = -2 * A.t()

@@ -191,14 +191,15 @@ def test_enet_influence_matches_reference():
     rho = torch.tensor([[0.05, 0.01]], device=DEV)
     xg, Yg, Sg, nh = ops.ext().enet_lbfgs_solve(Ab, yb, rho, 20, 10, 7)
     pen = torch.zeros(1, device=DEV)
-    EE, reward = ops.ext().enet_influence(Ab, yb, xg, Yg, Sg, nh, pen)
+    EE, reward = ops.ext().enet_influence(Ab, yb, xg, Yg, Sg, nh, pen,
+                                          rho)
     torch.cuda.synchronize()
 
     # CPU oracle on the SAME curvature pairs (from the GPU solve)
     k = int(nh[0])
     Yc = Yg[0, :k].cpu()
     Sc = Sg[0, :k].cpu()
-    EE_ref = enet_ops.influence_eigs_reference(A, Yc, Sc)
+    EE_ref = enet_ops.influence_eigs_reference(A, Yc, Sc, rho1=0.05)
     assert torch.allclose(EE[0].cpu(), EE_ref, rtol=1e-2, atol=1e-3), \
         f"max err {(EE[0].cpu()-EE_ref).abs().max()}"
 
@@ -383,14 +384,15 @@ def test_enet_solver_batched_matches_single():
     xb, Yb, Sb, nhb = ops.ext().enet_lbfgs_solve(
         A.contiguous(), y.contiguous(), rho.contiguous(), 20, 10, 7)
     EEb, rb = ops.ext().enet_influence(A.contiguous(), y.contiguous(),
-                                       xb, Yb, Sb, nhb, pen)
+                                       xb, Yb, Sb, nhb, pen,
+                                       rho.contiguous())
     for e in range(E):
         x1, Y1, S1, nh1 = ops.ext().enet_lbfgs_solve(
             A[e:e + 1].contiguous(), y[e:e + 1].contiguous(),
             rho[e:e + 1].contiguous(), 20, 10, 7)
         EE1, r1 = ops.ext().enet_influence(
             A[e:e + 1].contiguous(), y[e:e + 1].contiguous(),
-            x1, Y1, S1, nh1, pen[e:e + 1])
+            x1, Y1, S1, nh1, pen[e:e + 1], rho[e:e + 1].contiguous())
         assert torch.equal(xb[e], x1[0]), e
         assert torch.equal(EEb[e], EE1[0]), e
         assert torch.equal(rb[e], r1[0]), e
