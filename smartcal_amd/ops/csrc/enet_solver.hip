@@ -507,26 +507,33 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
   for (int m = 0; m < 32; ++m)
     qc[m] = (lane < N && m < M) ? -2.f * A[lane * M + m] : 0.f;
 
-  // Degenerate-pair filter AT THE ANALYTIC BOUND: the true Hessian is
-  // 2(A^T A + rho1 I) and the L1 subgradient difference is monotone, so
-  // every genuine curvature pair has ys >= 2 rho1 ss. Pairs below
-  // rho1 ss (half the floor) are line-search noise whose 1/ys blows the
-  // two-loop up — with the old fixed 1e-6 cutoff, noise pairs in
-  // (1e-6, rho1) ss still inflated H^-1 ~1000x over the worst genuine
-  // pair and produced the huge negative min(EE)/max(EE) rewards the
-  // round-2 curve audit caught. ro = 0 no-ops a pair in both loops. The
-  // CPU oracle (`ops/enet.py::influence_eigs_reference`) applies the
-  // same filter.
-  const float thr = fmaxf(1e-6f, rhog[env * 2]);
+  // Degenerate-pair filter AT THE ANALYTIC BAND. The smooth Hessian is
+  // 2(A^T A + rho1 I) with ||A||_F = 1, so every genuine pair y = H s
+  // satisfies BOTH  ys >= 2 rho1 ss  (monotone L1 subgradient included)
+  // AND  yy <= 4 (1+rho1)^2 ss.  Line-search noise violates one side or
+  // the other: tiny-ys pairs inflate the two-loop via 1/ys, and pairs
+  // whose y is dominated by L1 sign jumps over a minuscule step have
+  // ys/ss up to ~3e5 (measured, gpurun_out/explosion_debug.json) —
+  // their (I - rho s y^T) factors carry norm 1/cos(y,s) and blow
+  // H^{-1}'s spectrum up, producing the huge negative min(EE)/max(EE)
+  // rewards the round-2 curve audit caught. Filter: rho1 ss < ys AND
+  // yy < 8 (1+rho1)^2 ss (2x margin each side); ro = 0 no-ops a pair in
+  // both loops. With every pair rejected H^{-1} = I, so EE stays in
+  // [1 - 2 sigma_max^2, 1] — bounded. The CPU oracle
+  // (`ops/enet.py::influence_eigs_reference`) applies the same band.
+  const float rho1f = rhog[env * 2];
+  const float lo = fmaxf(1e-6f, rho1f);
+  const float hi = 8.f * (1.f + rho1f) * (1.f + rho1f);
   float ys = 1.f, yy = 1.f;
   for (int i = 0; i < nh; ++i) {
     float ysi = lds_dot(&Yv[i * M], &Sv[i * M], M);
     float ssi = lds_dot(&Sv[i * M], &Sv[i * M], M);
-    float r = (ysi > thr * ssi) ? 1.f / ysi : 0.f;
+    float yyi = lds_dot(&Yv[i * M], &Yv[i * M], M);
+    float r = (ysi > lo * ssi && yyi < hi * ssi) ? 1.f / ysi : 0.f;
     if (lane == 0) ro[i] = r;
     if (r != 0.f) {  // H_diag scale from the newest GOOD pair
       ys = ysi;
-      yy = lds_dot(&Yv[i * M], &Yv[i * M], M);
+      yy = yyi;
     }
   }
   __builtin_amdgcn_s_barrier();

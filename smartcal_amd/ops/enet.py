@@ -83,14 +83,19 @@ def influence_eigs_reference(A: torch.Tensor, Y: torch.Tensor,
     from ..autograd_tools import inv_hessian_mult_mat
 
     if Y.shape[0] > 0:
-        # degenerate-pair filter at the analytic bound (same as the HIP
-        # influence kernel): the true Hessian 2(A^T A + rho1 I) and the
-        # monotone L1 subgradient give ys >= 2 rho1 ss for every genuine
-        # pair; pairs below rho1 ss are line-search noise whose 1/ys
-        # explodes the two-loop (and the min(EE)/max(EE) reward term)
+        # degenerate-pair filter at the analytic BAND (same as the HIP
+        # influence kernel): genuine pairs of the true Hessian
+        # 2(A^T A + rho1 I), ||A||_F = 1, satisfy ys >= 2 rho1 ss AND
+        # yy <= 4 (1+rho1)^2 ss. Pairs outside (2x margin each side) are
+        # line-search noise: tiny-ys pairs inflate the two-loop via
+        # 1/ys, huge-||y|| pairs (L1 sign jumps over a minuscule step,
+        # ys/ss up to ~3e5 measured) inflate it via 1/cos(y, s) — both
+        # explode the min(EE)/max(EE) reward term.
         ys = (Y * S).sum(-1)
         ss = (S * S).sum(-1)
-        good = ys > max(1e-6, float(rho1)) * ss
+        yy = (Y * Y).sum(-1)
+        hi = 8.0 * (1.0 + float(rho1)) ** 2
+        good = (ys > max(1e-6, float(rho1)) * ss) & (yy < hi * ss)
         Y, S = Y[good], S[good]
     Q = -2.0 * A.t().contiguous()
     mm = inv_hessian_mult_mat(Y, S, Q)
