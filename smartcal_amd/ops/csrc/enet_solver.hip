@@ -635,7 +635,12 @@ extern "C" __global__ __launch_bounds__(64) void enet_influence_kernel(
       }
     }
   }
-  if (lane < N) ev[lane] = B[lane * N + lane] + 1.f;
+  // Projection onto the analytically feasible interval: the TRUE
+  // H^{-1} <= I/(2 rho1) and sigma_max(A) <= ||A||_F = 1 bound
+  // eig(B) >= -1/rho1, i.e. EE >= 1 - 1/rho1; eigenvalues below are
+  // quasi-Newton approximation artifacts (the rare hint-arm tail).
+  const float ee_floor = 1.f - 1.f / fmaxf(rho1f, 1e-6f);
+  if (lane < N) ev[lane] = fmaxf(B[lane * N + lane] + 1.f, ee_floor);
 
   // odd-even transposition sort ascending (N <= 32)
   for (int phase = 0; phase < N; ++phase) {

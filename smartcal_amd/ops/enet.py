@@ -102,7 +102,10 @@ def influence_eigs_reference(A: torch.Tensor, Y: torch.Tensor,
     B = A @ mm
     Bs = 0.5 * (B + B.t())
     ev = torch.linalg.eigvalsh(Bs.cpu()).to(A.device)
-    return ev + 1.0
+    # projection onto the analytically feasible interval: the TRUE
+    # H^{-1} <= I/(2 rho1) and sigma_max(A) <= ||A||_F = 1 bound
+    # eig(B) >= -1/rho1; values below are quasi-Newton artifacts
+    return (ev + 1.0).clamp(min=1.0 - 1.0 / max(float(rho1), 1e-6))
 
 
 def solve_and_influence_device(A: torch.Tensor, y: torch.Tensor,

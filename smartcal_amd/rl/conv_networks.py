@@ -127,7 +127,9 @@ class SACActorCNN(nn.Module):
 
     def sample_normal(self, img, meta, reparameterize: bool = True):
         mu, sigma = self.forward(img, meta)
-        dist = Normal(mu, sigma)
+        # validate_args syncs the host (.all() on a GPU bool) —
+        # illegal inside hipGraph capture
+        dist = Normal(mu, sigma, validate_args=False)
         raw = dist.rsample() if reparameterize else dist.sample()
         action = torch.tanh(raw) * self.max_action
         log_probs = dist.log_prob(raw) \
