@@ -327,20 +327,25 @@ class Agent:
         self._g_hint = torch.zeros(B, self.n_actions, device=dev)
         self._fill_static()
         torch.cuda.synchronize()
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
-            for _ in range(3):
+        # MIOpen batch-norm does find-db lookups/allocations at call time
+        # that break stream capture — force the native implementation for
+        # the captured body (the graph then always replays native kernels)
+        with torch.backends.cudnn.flags(enabled=False):
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    self._learn_body(self._g_img, self._g_meta,
+                                     self._g_action, self._g_reward,
+                                     self._g_img_, self._g_meta_,
+                                     self._g_done, self._g_hint)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
                 self._learn_body(self._g_img, self._g_meta, self._g_action,
                                  self._g_reward, self._g_img_,
                                  self._g_meta_, self._g_done, self._g_hint)
-        torch.cuda.current_stream().wait_stream(s)
-        torch.cuda.synchronize()
-        g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            self._learn_body(self._g_img, self._g_meta, self._g_action,
-                             self._g_reward, self._g_img_, self._g_meta_,
-                             self._g_done, self._g_hint)
         self._graph = g
 
     def disable_cuda_graph(self):
