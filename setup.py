@@ -30,6 +30,7 @@ sources = [str(CSRC / f) for f in [
     "fused_linear_bf16.hip",
     "attention.hip",
     "cgemm.hip",
+    "coherency.hip",
 ]]
 
 setup(

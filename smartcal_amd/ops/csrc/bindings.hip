@@ -74,6 +74,9 @@ extern "C" __global__ void attn_bwd_kernel(const float*, const float*,
 extern "C" __global__ void cgemm_nn_bcast_kernel(const float*, const float*,
                                                  float*, int, int, int, int,
                                                  int);
+extern "C" __global__ void coherency_kernel(const double*, const double*,
+                                            const int*, float*, double,
+                                            int, int);
 extern "C" __global__ void per_update_kernel(float*, const long*,
                                              const float*, int, float, float,
                                              float);
@@ -584,6 +587,29 @@ at::Tensor cgemm_nn_bcast(const at::Tensor& A, const at::Tensor& B,
   return C;
 }
 
+// Whole-sky coherency prediction in one launch (N8).
+at::Tensor coherency_predict(const at::Tensor& uvw_scaled,
+                             const at::Tensor& src, const at::Tensor& off,
+                             double fdelta) {
+  TORCH_CHECK(uvw_scaled.is_cuda()
+              && uvw_scaled.scalar_type() == at::kDouble
+              && uvw_scaled.is_contiguous(), "uvw must be f64 contiguous");
+  TORCH_CHECK(src.scalar_type() == at::kDouble && src.is_contiguous(),
+              "src table must be f64 contiguous");
+  TORCH_CHECK(off.scalar_type() == at::kInt, "offsets must be int32");
+  const int T = uvw_scaled.size(0);
+  const int K = off.numel() - 1;
+  TORCH_CHECK(src.size(1) == 11, "src table must be (S, 11)");
+  auto C = at::zeros({K, T, 4},
+                     uvw_scaled.options().dtype(at::kComplexFloat));
+  dim3 grid((T + 255) / 256, K);
+  hipLaunchKernelGGL(coherency_kernel, grid, dim3(256), 0, stream(),
+                     uvw_scaled.data_ptr<double>(), src.data_ptr<double>(),
+                     off.data_ptr<int>(),
+                     reinterpret_cast<float*>(C.data_ptr()), fdelta, T, K);
+  return C;
+}
+
 // Fused small-sequence attention: O = softmax(Q K^T / sqrt(dh)) V in one
 // launch per call; A (softmax) returned for backward/inspection.
 std::tuple<at::Tensor, at::Tensor> attn_fwd(const at::Tensor& Q,
@@ -654,4 +680,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("cgemm_nn_bcast", &cgemm_nn_bcast);
+  m.def("coherency_predict", &coherency_predict);
 }

@@ -61,6 +61,14 @@ def predict_coherencies_uvw(sky: SkyModel, clusters: ClusterSet,
     w = uvw[:, 2].double() * scale
     fdelta = (smear_bw / freq) if smear_bw is not None else None
 
+    from ..ops import use_hip
+    if use_hip(uvw.float() if uvw.is_cuda else uvw):
+        # ONE kernel launch for the whole sky (ops/csrc/coherency.hip);
+        # the torch composition below is the CPU oracle
+        return _predict_hip(sky, clusters, torch.stack((u, v, w), dim=1)
+                            .contiguous(), freq, ra0, dec0,
+                            fdelta if fdelta is not None else 0.0)
+
     K = len(clusters)
     C = torch.zeros((K, T, 4), dtype=torch.complex64, device=device)
     for ck, cluster in enumerate(clusters):
@@ -102,6 +110,48 @@ def predict_coherencies_uvw(sky: SkyModel, clusters: ClusterSet,
         C[ck, :, 0] = acc.to(torch.complex64)
         C[ck, :, 3] = C[ck, :, 0]
     return C
+
+
+def _predict_hip(sky, clusters, uvw_scaled, freq, ra0, dec0, fdelta):
+    """Build the per-source coefficient table and run the HIP kernel.
+
+    Table row: [l, m, n, flux, gflag, gu1, gv1, gw1, gu2, gv2, gw2] —
+    the Gaussian projected-envelope rotation is linear in (u, v, w), so
+    it collapses to 6 per-source coefficients (see coherency.hip)."""
+    from ..ops import ext
+
+    device = uvw_scaled.device
+    rows = []
+    offs = [0]
+    for cluster in clusters:
+        lmn, flux, eX2, eY2, eP, gflag = _cluster_source_tensors(
+            sky, cluster, ra0, dec0, freq, device)
+        S = lmn.shape[0]
+        tab = torch.zeros(S, 11, dtype=torch.float64, device=device)
+        tab[:, 0:3] = lmn
+        tab[:, 3] = flux
+        g = gflag.to(torch.bool)
+        if bool(g.any()):
+            gi = torch.nonzero(g, as_tuple=True)[0]
+            ll, mm, nn = lmn[gi, 0], lmn[gi, 1], lmn[gi, 2]
+            phi = -torch.acos(torch.clamp(nn, -1.0, 1.0))
+            xi = -torch.atan2(-ll, mm)
+            cxi, sxi = torch.cos(xi), torch.sin(xi)
+            cphi, sphi = torch.cos(phi), torch.sin(phi)
+            cpa, spa = torch.cos(eP[gi]), torch.sin(eP[gi])
+            # uup = (cxi)u + (-cphi sxi)v + (sphi sxi)w; vvp likewise
+            up = torch.stack((cxi, -cphi * sxi, sphi * sxi), dim=1)
+            vp = torch.stack((sxi, cphi * cxi, -sphi * cxi), dim=1)
+            tab[gi, 4] = 1.0
+            tab[gi, 5:8] = 2.0 * eX2[gi, None] * (cpa[:, None] * up
+                                                  - spa[:, None] * vp)
+            tab[gi, 8:11] = 2.0 * eY2[gi, None] * (spa[:, None] * up
+                                                   + cpa[:, None] * vp)
+        rows.append(tab)
+        offs.append(offs[-1] + S)
+    src = torch.cat(rows).contiguous()
+    off = torch.tensor(offs, dtype=torch.int32, device=device)
+    return ext().coherency_predict(uvw_scaled, src, off, float(fdelta))
 
 
 def predict_coherencies(sky: SkyModel, clusters: ClusterSet,
