@@ -162,8 +162,14 @@ class SACActorTransformer(nn.Module):
         mu, sigma = self.forward(img, meta)
         # validate_args syncs the host (.all() on a GPU bool) —
         # illegal inside hipGraph capture
+        # Normal kept for log_prob only (elementwise, capture-safe);
+        # the DRAW is explicit reparameterization — the two-tensor
+        # torch.normal(loc, scale) that rsample()/sample() lower to is
+        # hipErrorStreamCaptureUnsupported on ROCm, randn_like is fine
         dist = Normal(mu, sigma, validate_args=False)
-        raw = dist.rsample() if reparameterize else dist.sample()
+        raw = mu + sigma * torch.randn_like(mu)
+        if not reparameterize:
+            raw = raw.detach()
         action = torch.tanh(raw) * self.max_action
         log_probs = dist.log_prob(raw) \
             - torch.log(1 - action.pow(2) + self.reparam_noise)
