@@ -31,6 +31,7 @@ sources = [str(CSRC / f) for f in [
     "attention.hip",
     "cgemm.hip",
     "coherency.hip",
+    "hessianres.hip",
 ]]
 
 setup(

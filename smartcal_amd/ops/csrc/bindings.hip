@@ -77,6 +77,10 @@ extern "C" __global__ void cgemm_nn_bcast_kernel(const float*, const float*,
 extern "C" __global__ void coherency_kernel(const double*, const double*,
                                             const int*, float*, double,
                                             int, int);
+struct c32h2 { float x, y; };
+extern "C" __global__ void hessianres_kernel(
+    const c32h2*, const c32h2*, const c32h2*, const int*, const int*,
+    c32h2*, int, int, int, int);
 extern "C" __global__ void per_update_kernel(float*, const long*,
                                              const float*, int, float, float,
                                              float);
@@ -587,6 +591,30 @@ at::Tensor cgemm_nn_bcast(const at::Tensor& A, const at::Tensor& B,
   return C;
 }
 
+// Calibration Hessian assembly in one launch (N9).
+at::Tensor hessianres(const at::Tensor& C, const at::Tensor& R,
+                      const at::Tensor& J, const at::Tensor& p_idx,
+                      const at::Tensor& q_idx, int64_t N) {
+  TORCH_CHECK(C.is_cuda() && C.scalar_type() == at::kComplexFloat
+              && C.is_contiguous(), "C must be contiguous cfloat");
+  TORCH_CHECK(R.is_contiguous() && J.is_contiguous(), "contiguous");
+  TORCH_CHECK(p_idx.scalar_type() == at::kInt, "p_idx int32");
+  const int K = C.size(0);
+  const int S = C.size(1);
+  const int B = N * (N - 1) / 2;
+  const int T = S / B;
+  auto H = at::zeros({K, 4 * N, 4 * N}, C.options());
+  dim3 grid((B + 3) / 4, K);
+  hipLaunchKernelGGL(hessianres_kernel, grid, dim3(64), 0, stream(),
+                     reinterpret_cast<const c32h2*>(C.data_ptr()),
+                     reinterpret_cast<const c32h2*>(R.data_ptr()),
+                     reinterpret_cast<const c32h2*>(J.data_ptr()),
+                     p_idx.data_ptr<int>(), q_idx.data_ptr<int>(),
+                     reinterpret_cast<c32h2*>(H.data_ptr()),
+                     K, (int)N, B, T);
+  return H;
+}
+
 // Whole-sky coherency prediction in one launch (N8).
 at::Tensor coherency_predict(const at::Tensor& uvw_scaled,
                              const at::Tensor& src, const at::Tensor& off,
@@ -681,4 +709,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("cgemm_nn_bcast", &cgemm_nn_bcast);
   m.def("coherency_predict", &coherency_predict);
+  m.def("hessianres", &hessianres);
 }

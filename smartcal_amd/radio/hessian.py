@@ -81,6 +81,15 @@ def hessianres(R: torch.Tensor, C: torch.Tensor, J: torch.Tensor,
     B = N * (N - 1) // 2
     T = S // B
     p_idx, q_idx = baseline_pq(N, C.device)
+    from ..ops import use_hip
+    if use_hip(R.real):
+        # single-launch assembly kernel (ops/csrc/hessianres.hip); the
+        # torch composition below is the CPU oracle
+        from ..ops import ext
+        return ext().hessianres(
+            C.contiguous(), R.contiguous(), J.contiguous(),
+            p_idx.to(torch.int32).contiguous(),
+            q_idx.to(torch.int32).contiguous(), N)
     Res = R.reshape(S, 2, 2)
     Ci = _c22(C)                                          # (K,S,2,2)
 

@@ -333,6 +333,7 @@ class Agent:
         self._g_hint = torch.zeros(B, self.n_actions, device=dev)
         self._fill_static()
         torch.cuda.synchronize()
+        counter0 = self.learn_counter   # warmup/capture must not count
         # MIOpen batch-norm does find-db lookups/allocations at call time
         # that break stream capture — force the native implementation for
         # the captured body (the graph then always replays native kernels)
@@ -352,6 +353,7 @@ class Agent:
                 self._learn_body(self._g_img, self._g_meta, self._g_action,
                                  self._g_reward, self._g_img_,
                                  self._g_meta_, self._g_done, self._g_hint)
+        self.learn_counter = counter0
         self._graph = g
 
     def disable_cuda_graph(self):
