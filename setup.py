@@ -32,6 +32,7 @@ sources = [str(CSRC / f) for f in [
     "cgemm.hip",
     "coherency.hip",
     "hessianres.hip",
+    "two_loop.hip",
 ]]
 
 setup(

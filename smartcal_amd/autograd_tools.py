@@ -101,6 +101,16 @@ def inv_hessian_mult_mat(Y: torch.Tensor, S: torch.Tensor,
     ro = 1.0 / (Y * S).sum(dim=1)          # (h,)
     ys = float(Y[-1].dot(S[-1]))
     yy = float(Y[-1].dot(Y[-1]))
+    from .ops import use_hip
+    if use_hip(Q) and h <= 16:
+        # whole recursion for all RHS columns in ONE launch
+        # (ops/csrc/two_loop.hip — N6/N7); torch composition below is
+        # the CPU oracle
+        from .ops import ext
+        R = ext().two_loop_apply(Y.contiguous(), S.contiguous(),
+                                 Q.t().contiguous(), ro.contiguous(),
+                                 ys / yy)
+        return R.t().contiguous()
     Q = Q.clone()
     al = Q.new_zeros(h, Q.shape[1])
     for i in range(h - 1, -1, -1):

@@ -81,6 +81,10 @@ struct c32h2 { float x, y; };
 extern "C" __global__ void hessianres_kernel(
     const c32h2*, const c32h2*, const c32h2*, const int*, const int*,
     c32h2*, int, int, int, int);
+extern "C" __global__ void two_loop_kernel(const float*, const float*,
+                                           const float*, float*,
+                                           const float*, float, int, long,
+                                           int);
 extern "C" __global__ void per_update_kernel(float*, const long*,
                                              const float*, int, float, float,
                                              float);
@@ -591,6 +595,27 @@ at::Tensor cgemm_nn_bcast(const at::Tensor& A, const at::Tensor& B,
   return C;
 }
 
+// General L-BFGS two-loop for m RHS rows in one launch (N6/N7).
+at::Tensor two_loop_apply(const at::Tensor& Y, const at::Tensor& S,
+                          const at::Tensor& Q, const at::Tensor& ro,
+                          double gamma) {
+  check_f32(Y, "Y");
+  check_f32(S, "S");
+  check_f32(Q, "Q");
+  check_f32(ro, "ro");
+  const int h = Y.size(0);
+  const long n = Y.size(1);
+  const int m = Q.size(0);
+  TORCH_CHECK(h <= 16, "two_loop: history <= 16");
+  TORCH_CHECK(Q.size(1) == n && S.sizes() == Y.sizes(), "shape mismatch");
+  auto R = at::empty_like(Q);
+  hipLaunchKernelGGL(two_loop_kernel, dim3(m), dim3(256), 0, stream(),
+                     Y.data_ptr<float>(), S.data_ptr<float>(),
+                     Q.data_ptr<float>(), R.data_ptr<float>(),
+                     ro.data_ptr<float>(), (float)gamma, h, n, m);
+  return R;
+}
+
 // Calibration Hessian assembly in one launch (N9).
 at::Tensor hessianres(const at::Tensor& C, const at::Tensor& R,
                       const at::Tensor& J, const at::Tensor& p_idx,
@@ -710,4 +735,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cgemm_nn_bcast", &cgemm_nn_bcast);
   m.def("coherency_predict", &coherency_predict);
   m.def("hessianres", &hessianres);
+  m.def("two_loop_apply", &two_loop_apply);
 }

@@ -294,15 +294,31 @@ class LBFGSNew(Optimizer):
             else:
                 # two-loop recursion (reference lbfgsnew.py:637-651)
                 num_old = len(old_dirs)
-                al = [None] * num_old
-                q = flat_grad.neg()
-                for i in range(num_old - 1, -1, -1):
-                    al[i] = float(old_stps[i].dot(q)) * float(ro[i])
-                    q.add_(old_dirs[i], alpha=-al[i])
-                d = q.mul(H_diag)
-                for i in range(num_old):
-                    be_i = float(old_dirs[i].dot(d)) * float(ro[i])
-                    d.add_(old_stps[i], alpha=al[i] - be_i)
+                from ..ops import use_hip
+                if use_hip(flat_grad) and num_old <= 16:
+                    # whole recursion in ONE kernel launch
+                    # (ops/csrc/two_loop.hip) — the host composition
+                    # below syncs once per history entry per dot
+                    from ..ops import ext
+                    Y = torch.stack(list(old_dirs))
+                    S = torch.stack(list(old_stps))
+                    rov = torch.stack([torch.as_tensor(
+                        r, device=flat_grad.device).reshape(())
+                        for r in ro]).to(torch.float32)
+                    d = ext().two_loop_apply(
+                        Y.contiguous(), S.contiguous(),
+                        flat_grad.neg().reshape(1, -1).contiguous(),
+                        rov.contiguous(), float(H_diag)).reshape(-1)
+                else:
+                    al = [None] * num_old
+                    q = flat_grad.neg()
+                    for i in range(num_old - 1, -1, -1):
+                        al[i] = float(old_stps[i].dot(q)) * float(ro[i])
+                        q.add_(old_dirs[i], alpha=-al[i])
+                    d = q.mul(H_diag)
+                    for i in range(num_old):
+                        be_i = float(old_dirs[i].dot(d)) * float(ro[i])
+                        d.add_(old_stps[i], alpha=al[i] - be_i)
 
             if prev_flat_grad is None:
                 prev_flat_grad = flat_grad.clone(memory_format=torch.contiguous_format)
