@@ -33,6 +33,7 @@ sources = [str(CSRC / f) for f in [
     "coherency.hip",
     "hessianres.hip",
     "two_loop.hip",
+    "gather_sum.hip",
 ]]
 
 setup(

@@ -85,6 +85,9 @@ extern "C" __global__ void two_loop_kernel(const float*, const float*,
                                            const float*, float*,
                                            const float*, float, int, long,
                                            int);
+extern "C" __global__ void gather_sum_kernel(const c32b*, const long*,
+                                             c32b*, int, int, long, int,
+                                             int);
 extern "C" __global__ void per_update_kernel(float*, const long*,
                                              const float*, int, float, float,
                                              float);
@@ -595,6 +598,26 @@ at::Tensor cgemm_nn_bcast(const at::Tensor& A, const at::Tensor& B,
   return C;
 }
 
+// Fused gather + segment-sum (the ALS per-station reduction).
+at::Tensor gather_sum(const at::Tensor& in, const at::Tensor& gidx) {
+  TORCH_CHECK(in.is_cuda() && in.scalar_type() == at::kComplexFloat
+              && in.is_contiguous(), "in must be contiguous cfloat");
+  TORCH_CHECK(gidx.scalar_type() == at::kLong && gidx.is_contiguous(),
+              "gidx must be contiguous int64");
+  const int F = in.size(0), X = in.size(1);
+  const long S = in.size(2);
+  const int G = gidx.size(0), Cnt = gidx.size(1);
+  auto out = at::empty({F, X, G}, in.options());
+  const long total = (long)F * X * G;
+  hipLaunchKernelGGL(gather_sum_kernel,
+                     dim3((unsigned)((total + 3) / 4)), dim3(256), 0,
+                     stream(), reinterpret_cast<const c32b*>(in.data_ptr()),
+                     gidx.data_ptr<long>(),
+                     reinterpret_cast<c32b*>(out.data_ptr()), F, X, S, G,
+                     Cnt);
+  return out;
+}
+
 // General L-BFGS two-loop for m RHS rows in one launch (N6/N7).
 at::Tensor two_loop_apply(const at::Tensor& Y, const at::Tensor& S,
                           const at::Tensor& Q, const at::Tensor& ro,
@@ -736,4 +759,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("coherency_predict", &coherency_predict);
   m.def("hessianres", &hessianres);
   m.def("two_loop_apply", &two_loop_apply);
+  m.def("gather_sum", &gather_sum);
 }

@@ -145,12 +145,23 @@ def _solve_sweeps(data22, C22, J, rho_t, prox_target, p_idx, q_idx, N,
                 .permute(0, 2, 1)
         # reduce into (F,Ts,N,…) by interval and station via the
         # precomputed gather plan (see above); cat layout is (F, X, 2TB)
-        rhs = rhs_cat[:, :, gidx.reshape(-1)] \
-            .reshape(F, 2 * 2 * K, Ts * N, Cnt).sum(dim=3) \
-            .permute(0, 2, 1).reshape(F, Ts, N, 2, 2 * K)
-        nm = nm_cat[:, :, gidx.reshape(-1)] \
-            .reshape(F, 2 * K * 2 * K, Ts * N, Cnt).sum(dim=3) \
-            .permute(0, 2, 1).reshape(F, Ts, N, 2 * K, 2 * K)
+        if use_kernel:
+            # fused gather+segment-sum (ops/csrc/gather_sum.hip): the
+            # torch composition below materializes the full gathered
+            # copy (~260 MB/sweep at LOFAR scale) before reducing
+            from ..ops import ext as _ext
+            gc = gidx.contiguous()
+            rhs = _ext().gather_sum(rhs_cat.contiguous(), gc) \
+                .permute(0, 2, 1).reshape(F, Ts, N, 2, 2 * K)
+            nm = _ext().gather_sum(nm_cat.contiguous(), gc) \
+                .permute(0, 2, 1).reshape(F, Ts, N, 2 * K, 2 * K)
+        else:
+            rhs = rhs_cat[:, :, gidx.reshape(-1)] \
+                .reshape(F, 2 * 2 * K, Ts * N, Cnt).sum(dim=3) \
+                .permute(0, 2, 1).reshape(F, Ts, N, 2, 2 * K)
+            nm = nm_cat[:, :, gidx.reshape(-1)] \
+                .reshape(F, 2 * K * 2 * K, Ts * N, Cnt).sum(dim=3) \
+                .permute(0, 2, 1).reshape(F, Ts, N, 2 * K, 2 * K)
         # ADMM prox: + diag(ρ_k I2) and + ρ_k F^k on the rhs
         if prox_target is not None:
             rho_blocks = torch.kron(

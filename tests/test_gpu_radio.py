@@ -303,3 +303,17 @@ def test_hessianres_kernel_matches_torch_oracle():
         scale = H_ref.abs().max().clamp(min=1e-6)
         err = (H_k - H_ref).abs().max() / scale
         assert float(err) < 1e-5, (N, T, K, float(err))
+
+
+@needs_gpu
+def test_gather_sum_kernel_matches_torch():
+    from smartcal_amd import ops
+    torch.manual_seed(0)
+    F, X, S, G, Cnt = 3, 24, 4000, 50, 80
+    inp = (torch.randn(F, X, S) + 1j * torch.randn(F, X, S)) \
+        .to(torch.complex64).cuda()
+    gidx = torch.randint(0, S, (G, Cnt)).long().cuda()
+    out = ops.ext().gather_sum(inp.contiguous(), gidx.contiguous())
+    ref = inp[:, :, gidx.reshape(-1)].reshape(F, X, G, Cnt).sum(dim=3)
+    scale = ref.abs().max().clamp(min=1.0)
+    assert float((out - ref).abs().max() / scale) < 1e-5
