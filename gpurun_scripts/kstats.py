@@ -39,7 +39,9 @@ def main():
     ndisp = sum(r[1] for r in rows)
     print(f"{'kernel':<72} {'calls':>6} {'tot_ms':>8} {'avg_us':>7} "
           f"{'pct':>5}")
-    for name, calls, ms, us in rows[:40]:
+    import os
+    nrows = int(os.environ.get("KSTATS_ROWS", "40"))
+    for name, calls, ms, us in rows[:nrows]:
         print(f"{name[:72]:<72} {calls:>6} {ms:>8.2f} {us:>7.1f} "
               f"{100 * ms / total:>5.1f}")
     print(f"TOTAL kernel-ms: {total:.1f} over {ndisp} dispatches")
