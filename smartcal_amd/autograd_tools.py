@@ -68,6 +68,14 @@ def inv_hessian_mult(opt, q: torch.Tensor) -> torch.Tensor:
     dirs, stps = _curvature_pairs(opt)
     if not dirs or not stps:
         return q
+    if q.is_cuda:
+        # one two_loop_kernel launch instead of a host sync per history
+        # entry per dot (the CPU composition below)
+        Y = torch.stack(list(dirs))
+        S = torch.stack(list(stps))
+        r = inv_hessian_mult_mat(Y, S, q.reshape(-1, 1))[:, 0]
+        q.copy_(r)     # reference contract: q is scratch, r returned
+        return r
     n = len(dirs)
     ys = dirs[-1].dot(stps[-1])
     yy = dirs[-1].dot(dirs[-1])
