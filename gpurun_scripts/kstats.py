@@ -39,7 +39,6 @@ def main():
     ndisp = sum(r[1] for r in rows)
     print(f"{'kernel':<72} {'calls':>6} {'tot_ms':>8} {'avg_us':>7} "
           f"{'pct':>5}")
-    import os
     nrows = int(os.environ.get("KSTATS_ROWS", "40"))
     for name, calls, ms, us in rows[:nrows]:
         print(f"{name[:72]:<72} {calls:>6} {ms:>8.2f} {us:>7.1f} "
