@@ -83,6 +83,10 @@ class ReplayBuffer:
         no per-transition host sync) — the VecENetEnv rollout path. Ring
         semantics identical to E sequential store_transition calls."""
         E = states.shape[0]
+        if E > self.mem_size:
+            raise ValueError(
+                f"store_batch of {E} > capacity {self.mem_size}: duplicate "
+                "ring slots in one write are order-undefined on GPU")
         idx = (torch.arange(E, device=self.device)
                + self.mem_cntr) % self.mem_size
         self.state_memory[idx] = states.detach().to(
