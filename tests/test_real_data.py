@@ -144,3 +144,42 @@ def test_average_visdata_time_freq(tmp_path):
               + vis.data[1, 0] + vis.data[1, B]) / 4
     torch.testing.assert_close(av.data[0, 0], expect)
     assert np.isclose(av.freqs[0], np.mean(vis.freqs))
+
+
+def test_fits_reads_f64_and_degenerate_axes(tmp_path):
+    """Radio imagers write BITPIX -64 and trailing NAXIS3/4 = 1."""
+    from smartcal_amd.radio import fits_io
+    img = np.arange(12, dtype=np.float64).reshape(3, 4)
+    # hand-build a 4-axis f64 FITS
+    cards = [
+        f"{'SIMPLE':<8}= {'T':>20}", f"{'BITPIX':<8}= {-64:>20d}",
+        f"{'NAXIS':<8}= {4:>20d}", f"{'NAXIS1':<8}= {4:>20d}",
+        f"{'NAXIS2':<8}= {3:>20d}", f"{'NAXIS3':<8}= {1:>20d}",
+        f"{'NAXIS4':<8}= {1:>20d}", "END"]
+    hdr = b"".join(c[:80].ljust(80).encode() for c in cards)
+    hdr += b" " * (-len(hdr) % 2880)
+    data = img.astype(">f8").tobytes()
+    data += b"\x00" * (-len(data) % 2880)
+    p = tmp_path / "f64.fits"
+    p.write_bytes(hdr + data)
+    back = fits_io.read_image(str(p))
+    np.testing.assert_allclose(back, img.astype(np.float32))
+
+
+def test_average_visdata_rejects_over_averaging(tmp_path):
+    from smartcal_amd.radio import ms_io
+    paths, vis = _fixture_obs(tmp_path, Nf=2)
+    merged = ms_io.observation_from_npz(paths)
+    with pytest.raises(ValueError, match="exceed"):
+        ms_io.average_visdata(merged, freq_factor=5)
+
+
+def test_merge_visdata_rejects_geometry_mismatch(tmp_path):
+    from smartcal_amd.radio import ms_io, sim
+    paths, vis = _fixture_obs(tmp_path, Nf=2)
+    a = ms_io.observation_from_npz(paths[:1])
+    bad = sim.VisData(uvw=a.uvw, freqs=a.freqs,
+                      data=a.data[:, : a.data.shape[1] // 2], N=a.N,
+                      ra0=a.ra0, dec0=a.dec0, Ts=a.Ts, Tdelta=a.Tdelta)
+    with pytest.raises(ValueError, match="mismatch"):
+        ms_io.merge_visdata([a, bad])
