@@ -265,6 +265,7 @@ class Agent:
         self.critic_1_opt.zero_grad()
         self.critic_2_opt.zero_grad()
         c_loss.backward()
+        self._fork.join()   # side-stream backwards must land first
         if self.grad_hook is not None:
             self.grad_hook([self.critic_1_fp, self.critic_2_fp])
         self.critic_1_opt.step()
@@ -287,6 +288,7 @@ class Agent:
                 + self.rho * gfun
         self.actor_opt.zero_grad()
         actor_loss.backward()
+        self._fork.join()   # side-stream backwards must land first
         if self.grad_hook is not None:
             self.grad_hook([self.actor_fp])
         self.actor_opt.step()

@@ -180,6 +180,7 @@ class Agent:
         self.critic_1_opt.zero_grad()
         self.critic_2_opt.zero_grad()
         c_loss.backward()
+        self._fork.join()   # side-stream backwards must land first
         if self.grad_hook is not None:
             self.grad_hook([self.critic_1_fp, self.critic_2_fp])
         self.critic_1_opt.step()
@@ -196,6 +197,7 @@ class Agent:
             actor_loss = -self.critic_1(img, meta, actions).mean()
             self.actor_opt.zero_grad()
             actor_loss.backward()
+            self._fork.join()   # side-stream backwards must land first
             if self.grad_hook is not None:
                 self.grad_hook([self.actor_fp])
             self.actor_opt.step()
@@ -216,6 +218,7 @@ class Agent:
                 actor_loss = aloss + penalty
                 self.actor_opt.zero_grad()
                 actor_loss.backward()
+                self._fork.join()   # side-stream backwards must land first
                 if self.grad_hook is not None:
                     self.grad_hook([self.actor_fp])
                 self.actor_opt.step()

@@ -48,3 +48,20 @@ class StreamFork:
                     if torch.is_tensor(t):
                         t.record_stream(cs)
         return tuple(results)
+
+    def join(self):
+        """Make the current stream wait for everything enqueued on the
+        side streams so far. Needed before a gradient all-reduce:
+        backward of a forward that ran on a side stream also runs there
+        (stream-aware autograd), and the direct-accumulate backward
+        kernels bypass AccumulateGrad's leaf-stream sync. Graph-capture
+        safe (event record/wait become dependency edges)."""
+        if self.device.type != "cuda" or not self._streams:
+            return
+        if not hasattr(self, "_sync_evs"):
+            self._sync_evs = []
+        while len(self._sync_evs) < len(self._streams):
+            self._sync_evs.append(torch.cuda.Event())
+        for ev, s in zip(self._sync_evs, self._streams):
+            ev.record(s)
+            ev.wait()
