@@ -20,6 +20,7 @@
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
 #define NWAVES 8
 #define NT_MAX 12          // N <= 16*NWAVES*NT_MAX = 1536
@@ -67,11 +68,18 @@ extern "C" __global__ __launch_bounds__(512) void fused_linear_bf16_fwd_kernel(
   __bf16* wsw = ws + wave * 16 * XS;
 
   for (int kk = 0; kk < K; kk += BK) {
-    for (int idx = tid; idx < 16 * BK; idx += 512) {
+    // 4 elements per thread, one packed 8-byte LDS store each
+    for (int idx = tid * 4; idx < 16 * BK; idx += 512 * 4) {
       const int r = idx / BK, c = idx % BK;
-      const int gr = row0 + r, gc = kk + c;
-      xs[r * XS + c] =
-          (gr < B && gc < K) ? (__bf16)X[(long)gr * K + gc] : (__bf16)0.f;
+      const int gr = row0 + r;
+      bf16x4 p;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int gc = kk + c + j;
+        p[j] = (gr < B && gc < K) ? (__bf16)X[(long)gr * K + gc]
+                                  : (__bf16)0.f;
+      }
+      *reinterpret_cast<bf16x4*>(&xs[r * XS + c]) = p;
     }
     __syncthreads();
     const int kmax = min(BK, K - kk);
@@ -101,10 +109,11 @@ extern "C" __global__ __launch_bounds__(512) void fused_linear_bf16_fwd_kernel(
 #pragma unroll
       for (int i = 0; i < WCH; ++i) {
         const int o = wc_row[i] * XS + wc_k[i];
-        wsw[o + 0] = (__bf16)vr[i].x;
-        wsw[o + 1] = (__bf16)vr[i].y;
-        wsw[o + 2] = (__bf16)vr[i].z;
-        wsw[o + 3] = (__bf16)vr[i].w;
+        // wc_k multiples of 4 and XS even => 8-byte aligned: one
+        // ds_write_b64 instead of four 2-byte stores
+        bf16x4 p = {(__bf16)vr[i].x, (__bf16)vr[i].y, (__bf16)vr[i].z,
+                    (__bf16)vr[i].w};
+        *reinterpret_cast<bf16x4*>(&wsw[o]) = p;
       }
     };
 
@@ -242,9 +251,9 @@ extern "C" __global__ __launch_bounds__(256) void gemm_bf16_nn_kernel(
     }
   };
   auto store_a = [&](int buf, const float4* v) {
-#pragma unroll
-    for (int j = 0; j < 4; ++j)
-      as[buf][st_r * XS + st_c + j] = (__bf16)(&v->x)[j];
+    // st_c is a multiple of 4 and XS is even: one packed 8-byte store
+    bf16x4 p = {(__bf16)v->x, (__bf16)v->y, (__bf16)v->z, (__bf16)v->w};
+    *reinterpret_cast<bf16x4*>(&as[buf][st_r * XS + st_c]) = p;
   };
   f32x4 acc = (f32x4){0.f, 0.f, 0.f, 0.f};
   float4 areg;
