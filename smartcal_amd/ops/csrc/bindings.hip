@@ -168,6 +168,8 @@ struct ChainArgs {
   int B;
 };
 extern "C" __global__ void mlp_chain_fwd_kernel(const float*, ChainArgs);
+extern "C" __global__ void mlp_chain_bf16_fwd_kernel(const float*,
+                                                     ChainArgs);
 
 // One kernel for a whole Linear(+LN)(+act) chain (≤4 layers, widths ≤512):
 // activations ping-pong in LDS; per-layer y/zhat/rstd still stored for
@@ -179,7 +181,7 @@ mlp_chain_fwd(const at::Tensor& x,
               const std::vector<at::Tensor>& bs,
               const std::vector<at::Tensor>& gammas,
               const std::vector<at::Tensor>& betas,
-              const std::vector<int64_t>& acts) {
+              const std::vector<int64_t>& acts, bool bf16 = false) {
   check_f32(x, "x");
   const int L = (int)Ws.size();
   TORCH_CHECK(L >= 1 && L <= 4, "chain supports 1..4 layers");
@@ -215,6 +217,23 @@ mlp_chain_fwd(const at::Tensor& x,
   // + row stats — ~139 KB of the 160 KB LDS (one block per CU; only
   // ceil(B/16) blocks exist anyway)
   const int XP = 577;
+  if (bf16) {
+    // bf16 W region: 16 waves x 16 x 72 bf16 rounded up to floats
+    const int lds_bytes =
+        (2 * 16 * XP + (16 * 16 * 72 + 1) / 2 + 16 * 16 * 2 + 16 * 2)
+        * (int)sizeof(float);
+    static bool attr_set_b = false;
+    if (!attr_set_b) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&mlp_chain_bf16_fwd_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+      attr_set_b = true;
+    }
+    hipLaunchKernelGGL(mlp_chain_bf16_fwd_kernel, dim3((B + 15) / 16),
+                       dim3(1024), lds_bytes, stream(),
+                       x.data_ptr<float>(), args);
+    return {ys, zhats, rstds};
+  }
   const int lds_bytes =
       (2 * 16 * XP + 16 * 16 * 65 + 16 * 16 * 2 + 16 * 2)
       * (int)sizeof(float);

@@ -334,3 +334,242 @@ extern "C" __global__ __launch_bounds__(256) void gemm_bf16_tn_kernel(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Whole-chain MLP forward, bf16 compute (the bf16-mode counterpart of
+// mlp_chain_fwd_kernel in fused_linear.hip): activations ping-pong in
+// fp32 LDS (cast to bf16 at fragment assembly — identical numerics to
+// the per-layer bf16 path, which casts on staging), W subtiles staged
+// as packed bf16, GEMMs on v_mfma_f32_16x16x32_bf16, fp32 LN epilogue.
+// ---------------------------------------------------------------------------
+
+#define CHAIN_MAX 4
+#define CNW 16          // waves per block (1024 threads)
+#define CNT 3           // max 16-col tiles per wave (16*16*3 = 768 >= 512)
+#define XMAX 576
+#define XP (XMAX + 1)
+#define WXS (BK + XPAD)  // bf16 W row stride
+
+struct ChainArgsB {
+  const float* W[CHAIN_MAX];
+  const float* bias[CHAIN_MAX];
+  const float* gamma[CHAIN_MAX];
+  const float* beta[CHAIN_MAX];
+  float* Y[CHAIN_MAX];
+  float* ZHAT[CHAIN_MAX];
+  float* RSTD[CHAIN_MAX];
+  int dims[CHAIN_MAX + 1];
+  int act[CHAIN_MAX];
+  int with_ln[CHAIN_MAX];
+  int L;
+  int B;
+};
+
+extern "C" __global__ __launch_bounds__(1024) void mlp_chain_bf16_fwd_kernel(
+    const float* __restrict__ X, ChainArgsB args) {
+  extern __shared__ float smem[];
+  float* buf0 = smem;                       // [16][XP] fp32 activations
+  float* buf1 = smem + 16 * XP;
+  __bf16* ws = reinterpret_cast<__bf16*>(smem + 2 * 16 * XP);
+  float* rowstat = smem + 2 * 16 * XP
+      + (CNW * 16 * WXS + 1) / 2;           // after bf16 region (in floats)
+  float* rowmv = rowstat + CNW * 16 * 2;
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+  const int row0 = blockIdx.x * 16;
+  const int B = args.B;
+
+  const int wc_row[WCH] = {(lane + 0 * WAVE) / (BK / 4),
+                           (lane + 1 * WAVE) / (BK / 4),
+                           (lane + 2 * WAVE) / (BK / 4),
+                           (lane + 3 * WAVE) / (BK / 4)};
+  const int wc_k[WCH] = {((lane + 0 * WAVE) % (BK / 4)) * 4,
+                         ((lane + 1 * WAVE) % (BK / 4)) * 4,
+                         ((lane + 2 * WAVE) % (BK / 4)) * 4,
+                         ((lane + 3 * WAVE) % (BK / 4)) * 4};
+  __bf16* wsw = ws + wave * 16 * WXS;
+
+  // stage the input (zero-padded to a BK boundary), fp32
+  {
+    const int K0 = args.dims[0];
+    const int Kpad = ((K0 + BK - 1) / BK) * BK;
+    for (int idx = tid; idx < 16 * Kpad; idx += 1024) {
+      const int r = idx / Kpad, c = idx % Kpad;
+      const int gr = row0 + r;
+      buf0[r * XP + c] = (gr < B && c < K0) ? X[(long)gr * K0 + c] : 0.f;
+    }
+  }
+  __syncthreads();
+
+  float* xb = buf0;
+  float* yb = buf1;
+
+  for (int l = 0; l < args.L; ++l) {
+    const int K = args.dims[l];
+    const int N = args.dims[l + 1];
+    const int ntiles = (N + 15) >> 4;
+    const float* W = args.W[l];
+    const int with_ln = args.with_ln[l];
+
+    f32x4 acc[CNT];
+#pragma unroll
+    for (int t = 0; t < CNT; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    for (int kk = 0; kk < K; kk += BK) {
+      const int kmax = min(BK, K - kk);
+      auto load_wtile = [&](int t, float4* vr) {
+#pragma unroll
+        for (int i = 0; i < WCH; ++i) {
+          const int ct = wave + CNW * t;
+          const int col = ct * 16 + wc_row[i];
+          const int k = wc_k[i];
+          float4 v = {0.f, 0.f, 0.f, 0.f};
+          if (ct < ntiles && col < N) {
+            const float* wrow = W + (long)col * K + kk;
+            if (k + 3 < kmax) {
+              v = *reinterpret_cast<const float4*>(wrow + k);
+            } else {
+              if (k + 0 < kmax) v.x = wrow[k + 0];
+              if (k + 1 < kmax) v.y = wrow[k + 1];
+              if (k + 2 < kmax) v.z = wrow[k + 2];
+              if (k + 3 < kmax) v.w = wrow[k + 3];
+            }
+          }
+          vr[i] = v;
+        }
+      };
+      auto write_wtile = [&](const float4* vr) {
+#pragma unroll
+        for (int i = 0; i < WCH; ++i) {
+          bf16x4 p = {(__bf16)vr[i].x, (__bf16)vr[i].y, (__bf16)vr[i].z,
+                      (__bf16)vr[i].w};
+          *reinterpret_cast<bf16x4*>(&wsw[wc_row[i] * WXS + wc_k[i]]) = p;
+        }
+      };
+      float4 wa[WCH], wb[WCH];
+      load_wtile(0, wa);
+#pragma unroll
+      for (int t = 0; t < CNT; ++t) {
+        const int ct = wave + CNW * t;
+        if (ct >= ntiles) break;
+        if (t % 2 == 0) {
+          write_wtile(wa);
+          if (t + 1 < CNT) load_wtile(t + 1, wb);
+        } else {
+          write_wtile(wb);
+          if (t + 1 < CNT) load_wtile(t + 1, wa);
+        }
+        // A fragments: fp32 activations -> bf16 at assembly
+        bf16x8 a0, a1, b0, b1;
+#pragma unroll
+        for (int r = 0; r < 8; ++r) {
+          a0[r] = (__bf16)xb[l15 * XP + kk + l4 * 8 + r];
+          a1[r] = (__bf16)xb[l15 * XP + kk + 32 + l4 * 8 + r];
+        }
+        b0 = *reinterpret_cast<const bf16x8*>(&wsw[l15 * WXS + l4 * 8]);
+        b1 = *reinterpret_cast<const bf16x8*>(
+            &wsw[l15 * WXS + 32 + l4 * 8]);
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[t],
+                                                         0, 0, 0);
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[t],
+                                                         0, 0, 0);
+      }
+    }
+
+    // ---- fp32 epilogue: bias + LN + act; write HBM + the LDS ping ----
+    const float* bias = args.bias[l];
+    const float* gamma = args.gamma[l];
+    const float* beta = args.beta[l];
+    float zrow[CNT][4];
+    float psum[4] = {0.f, 0.f, 0.f, 0.f}, psq[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int t = 0; t < CNT; ++t) {
+      const int ct = wave + CNW * t;
+      const int col = ct * 16 + l15;
+      const bool colv = (ct < ntiles) && (col < N);
+      const float bv = (colv && bias) ? bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float z = colv ? acc[t][r] + bv : 0.f;
+        zrow[t][r] = z;
+        psum[r] += z;
+        psq[r] += z * z;
+      }
+    }
+    if (with_ln) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        psum[r] = group16_sum(psum[r]);
+        psq[r] = group16_sum(psq[r]);
+      }
+      if (l15 == 0) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          rowstat[(wave * 16 + l4 * 4 + r) * 2 + 0] = psum[r];
+          rowstat[(wave * 16 + l4 * 4 + r) * 2 + 1] = psq[r];
+        }
+      }
+      __syncthreads();
+      if (tid < 16) {
+        float sa = 0.f, q = 0.f;
+#pragma unroll
+        for (int w = 0; w < CNW; ++w) {
+          sa += rowstat[(w * 16 + tid) * 2 + 0];
+          q += rowstat[(w * 16 + tid) * 2 + 1];
+        }
+        const float mean = sa / N;
+        float var = q / N - mean * mean;
+        const float rstd = rsqrtf(fmaxf(var, 0.f) + 1e-5f);
+        rowmv[tid * 2 + 0] = mean;
+        rowmv[tid * 2 + 1] = rstd;
+        const int grow = row0 + tid;
+        if (grow < B && args.RSTD[l]) args.RSTD[l][grow] = rstd;
+      }
+      __syncthreads();
+    }
+    {
+      const int Npad = ((N + BK - 1) / BK) * BK;
+      const int padw = Npad - N;
+      if (padw > 0) {
+        for (int idx = tid; idx < 16 * padw; idx += 1024) {
+          const int r = idx / padw, c = N + idx % padw;
+          yb[r * XP + c] = 0.f;
+        }
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int lrow = l4 * 4 + r;
+      const int grow = row0 + lrow;
+      float mean = 0.f, rstd = 1.f;
+      if (with_ln) {
+        mean = rowmv[lrow * 2 + 0];
+        rstd = rowmv[lrow * 2 + 1];
+      }
+#pragma unroll
+      for (int t = 0; t < CNT; ++t) {
+        const int ct = wave + CNW * t;
+        const int col = ct * 16 + l15;
+        if (ct < ntiles && col < N) {
+          float z = zrow[t][r];
+          float out;
+          if (with_ln) {
+            const float zh = (z - mean) * rstd;
+            if (grow < B) args.ZHAT[l][(long)grow * N + col] = zh;
+            out = apply_act(gamma[col] * zh + beta[col], args.act[l]);
+          } else {
+            out = apply_act(z, args.act[l]);
+          }
+          yb[lrow * XP + col] = out;
+          if (grow < B) args.Y[l][(long)grow * N + col] = out;
+        }
+      }
+    }
+    __syncthreads();
+    float* tmp = xb; xb = yb; yb = tmp;
+  }
+}

@@ -220,12 +220,14 @@ class _FusedChainFn(torch.autograd.Function):
         bs = [params[4 * l + 1] for l in range(L)]
         gs = [params[4 * l + 2] for l in range(L)]
         bes = [params[4 * l + 3] for l in range(L)]
+        bf16 = _COMPUTE_DTYPE == "bf16"
         ys, zhats, rstds = ext().mlp_chain_fwd(
-            x, Ws, bs, gs, bes, [int(a) for a in acts])
+            x, Ws, bs, gs, bes, [int(a) for a in acts], bf16)
         ctx.save_for_backward(x, *ys, *zhats, *rstds)
         ctx.param_refs = params
         ctx.acts = acts
         ctx.L = L
+        ctx.bf16 = bf16
         return ys[-1]
 
     @staticmethod
@@ -252,10 +254,17 @@ class _FusedChainFn(torch.autograd.Function):
                     dy, ys[l], zhats[l], rstds[l], g, ctx.acts[l], True,
                     sb[:n], sb[n:2 * n])
             x_l = x if l == 0 else ys[l - 1]
-            if want_w:
-                ext().mfma_gemm_tn_bias_into(dz, x_l, _grad_view(W),
-                                             _grad_view(b))
-            dy = ext().mfma_gemm_nn(dz, W)
+            if getattr(ctx, "bf16", False):
+                if want_w:
+                    ext().mfma_gemm_tn_bias_into_bf16(dz, x_l,
+                                                      _grad_view(W),
+                                                      _grad_view(b))
+                dy = ext().mfma_gemm_nn_bf16(dz, W)
+            else:
+                if want_w:
+                    ext().mfma_gemm_tn_bias_into(dz, x_l, _grad_view(W),
+                                                 _grad_view(b))
+                dy = ext().mfma_gemm_nn(dz, W)
         return (dy, None) + (None,) * (4 * L)
 
 
@@ -266,7 +275,7 @@ def fused_chain(x: torch.Tensor, layers) -> torch.Tensor:
     squeeze = x.dim() == 1
     if squeeze:
         x = x.unsqueeze(0)
-    if use_hip(x) and len(layers) >= 2 and _COMPUTE_DTYPE == "fp32" \
+    if use_hip(x) and len(layers) >= 2 \
             and all(m.ln_weight is not None for m in layers):
         acts = tuple(_ACT_CODES[m.act] for m in layers)
         params = []

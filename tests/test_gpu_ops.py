@@ -720,3 +720,42 @@ def test_two_loop_kernel_matches_torch_oracle():
 
     xc, xg = run("cpu"), run("cuda")
     assert torch.allclose(xc, xg, atol=1e-3, rtol=1e-3)
+
+
+def test_bf16_chain_matches_per_layer_bf16():
+    """mlp_chain_bf16_fwd (whole chain, one launch) ≡ the per-layer bf16
+    path: identical casts and MFMA order, so results match tightly;
+    and the bf16-mode fused_chain trains (backward on bf16 GEMMs)."""
+    from smartcal_amd.ops import linear as linear_ops
+    from smartcal_amd.rl.networks import SACActorMLP
+    try:
+        linear_ops.set_compute_dtype("bf16")
+        torch.manual_seed(0)
+        x = torch.randn(64, 420, device=DEV)
+        Ws, bs, gs, bes = [], [], [], []
+        dims = [420, 512, 256, 128]
+        for i in range(3):
+            Ws.append(torch.randn(dims[i + 1], dims[i], device=DEV)
+                      / dims[i] ** 0.5)
+            bs.append(torch.randn(dims[i + 1], device=DEV))
+            gs.append(torch.rand(dims[i + 1], device=DEV) + 0.5)
+            bes.append(torch.randn(dims[i + 1], device=DEV))
+        ys, _, _ = ops.ext().mlp_chain_fwd(x, Ws, bs, gs, bes,
+                                           [1, 1, 1], True)
+        h = x
+        for i in range(3):
+            h, _, _ = ops.ext().fused_linear_bf16_fwd(
+                h.contiguous(), Ws[i], bs[i], gs[i], bes[i], 1, True)
+        assert torch.allclose(ys[-1], h, atol=1e-5, rtol=1e-5)
+        # end-to-end: a bf16-mode actor trains through the chain
+        torch.manual_seed(1)
+        actor = SACActorMLP(420, 2).cuda()
+        from smartcal_amd.utils.flatten import FlatParams, FusedAdam
+        fp = FlatParams(actor)
+        opt = FusedAdam(fp, lr=1e-3)
+        a, lp = actor.sample_normal(torch.randn(32, 420, device=DEV))
+        (lp.mean()).backward()
+        opt.step()
+        assert torch.isfinite(fp.flat).all()
+    finally:
+        linear_ops.set_compute_dtype("fp32")
