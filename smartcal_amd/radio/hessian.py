@@ -301,12 +301,20 @@ def dres_colmeans(C: torch.Tensor, J: torch.Tensor, N: int,
     # X = W · (H+εI)^{-1} : 4-RHS solve per k. H is Hermitian by
     # construction, so Cholesky (measured 1.85 ms vs 4.9 ms LU at
     # K=6, 4N=248) with an LU fallback for the indefinite corner.
+    # ROCm bug (probed on MI355X / ROCm 7.2, gpurun_scripts/
+    # chol384_probe.py): BATCHED complex cholesky_ex crashes with
+    # hipErrorLaunchFailure for n >= ~384 at batch > 1 (n=248 and any
+    # single-matrix size are fine) — gate the fast path to 4N <= 256
+    # and use the (everywhere-working) LU solve for larger arrays.
     Hh = H + _EPS * eye
-    L, info = torch.linalg.cholesky_ex(Hh)
-    if int(info.abs().sum()) == 0:
-        X = torch.cholesky_solve(Wmat.mH, L).mH                # (K,4,4N)
-    else:
+    if Hh.is_cuda and K > 1 and 4 * N > 256:
         X = torch.linalg.solve(Hh.mT, Wmat.mT).mT
+    else:
+        L, info = torch.linalg.cholesky_ex(Hh)
+        if int(info.abs().sum()) == 0:
+            X = torch.cholesky_solve(Wmat.mH, L).mH            # (K,4,4N)
+        else:
+            X = torch.linalg.solve(Hh.mT, Wmat.mT).mT
 
     # gather X columns at (h, p_idx[c], cc) and contract with M:
     # out[r,(k),a,c] = Σ_{h,cc} Xg[k,a,h,c,cc] · M[r,k,c,h,cc]
