@@ -317,3 +317,29 @@ def test_gather_sum_kernel_matches_torch():
     ref = inp[:, :, gidx.reshape(-1)].reshape(F, X, G, Cnt).sum(dim=3)
     scale = ref.abs().max().clamp(min=1.0)
     assert float((out - ref).abs().max() / scale) < 1e-5
+
+
+@needs_gpu
+def test_influence_core_large_array():
+    """Large-array influence core (N=96, 4N=384): regression for the
+    ROCm batched-complex-potrf crash (hessian.py gates Cholesky to
+    4N<=256; larger arrays take the LU path)."""
+    from smartcal_amd.radio import hessian as hs
+    rng = np.random.default_rng(0)
+    N, T, K = 96, 4, 3
+    B = N * (N - 1) // 2
+    S = B * T
+    C = torch.from_numpy((rng.standard_normal((K, S, 4))
+                          + 1j * rng.standard_normal((K, S, 4))
+                          ).astype(np.complex64)).cuda() * 0.1
+    J = torch.from_numpy((rng.standard_normal((K, 2 * N, 2))
+                          + 1j * rng.standard_normal((K, 2 * N, 2))
+                          ).astype(np.complex64)).cuda()
+    R = torch.from_numpy((rng.standard_normal((2 * S, 2))
+                          + 1j * rng.standard_normal((2 * S, 2))
+                          ).astype(np.complex64)).cuda()
+    H = hs.hessianres(R, C, J, N)
+    m = hs.dres_colmeans(C, J, N, H)
+    torch.cuda.synchronize()
+    assert m.shape == (8, 4, B)
+    assert torch.isfinite(m.real).all() and torch.isfinite(m.imag).all()
