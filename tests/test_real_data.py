@@ -183,3 +183,19 @@ def test_merge_visdata_rejects_geometry_mismatch(tmp_path):
                       ra0=a.ra0, dec0=a.dec0, Ts=a.Ts, Tdelta=a.Tdelta)
     with pytest.raises(ValueError, match="mismatch"):
         ms_io.merge_visdata([a, bad])
+
+
+def test_ms2npz_cli_reports_casacore_requirement(tmp_path):
+    """The converter CLI exists and fails with the bridge hint when
+    casacore is absent (this image)."""
+    try:
+        import casacore  # noqa: F401
+        pytest.skip("casacore installed")
+    except ImportError:
+        pass
+    r = subprocess.run(
+        [sys.executable, str(ROOT / "scripts/tools/ms2npz.py"),
+         str(tmp_path / "fake.ms")],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode != 0
+    assert "casacore" in (r.stderr + r.stdout)
